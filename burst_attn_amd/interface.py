@@ -1,0 +1,394 @@
+"""BurstAttention ring orchestration — public API and autograd layer.
+
+Drop-in for the reference's ``burst_attn/burst_attn_interface.py``:
+``burst_attn_func`` / ``burst_attn_func_striped`` keep the exact signature
+and torch.autograd semantics (reference ``:135-158`` / ``:109-132``; grads
+for q, k, v only, ``:398``).  Layout is the flash layout ``[B, S, N, H]``
+(class docstring ``:162-168``); ``process_group`` is a torch.distributed
+ProcessGroup (None → WORLD).
+
+MI355X-first redesign decisions (vs the reference):
+  * ONE local-tile backend — the gfx950 HIP kernel pair behind
+    ``tile.get_tile_provider()`` — instead of the reference's
+    cuda/triton/math dispatch (``:40-51``).  The ``flash`` argument is
+    accepted for signature compatibility; the math-path layout
+    (``flash`` not in {"cuda","triton"}; [B,N,S,D]) is not supported.
+  * dq/dk/dv accumulate in fp32 across ring rounds (the reference
+    accumulates in the input dtype, fp16) and are cast to the input dtype
+    at the end.
+  * ring payloads and round structure are otherwise identical —
+    forward rings {k, v} (``:214-242``); backward rings
+    {delta-or-o, grad_output, q, lse} plus a separate travelling-dq ring
+    with one extra final hop (``:291-396``).
+
+Causal load-balancing layouts (see oracle/partition.py):
+  * ``OpBurstAttn``     — zigzag: rank r holds global chunks [r] and
+    [2W-1-r]; per-round half-tile dispatch per ``:221-235`` (fwd) and
+    ``:303-367`` (bwd).
+  * ``OpBurstAttnStrip`` — striped: token t on rank t mod W; per-round
+    one-token shift per ``:459-475`` (fwd) and ``:557-585`` (bwd).
+"""
+
+import math
+
+import torch
+
+from .comm import Ring, get_rank, replicate
+from .log_helper import get_logger
+from .tile import get_tile_provider
+
+_logger = get_logger(__name__, level="WARN")
+
+__all__ = ["burst_attn_func", "burst_attn_func_striped", "OpBurstAttn", "OpBurstAttnStrip"]
+
+
+def get_partition_id(double_group, r):
+    """Which source-rank offset's K/V (fwd) or Q (bwd) is held at round r.
+
+    Single ring: offset r-1 (reference ``burst_attn_interface.py:20-37``;
+    the double-group formula is multi-node-only and lands with the double
+    ring)."""
+    return r - 1
+
+
+def _record_stream(*tensors):
+    """Stream hygiene for buffers swapped out while RCCL may still read
+    them (reference ``burst_utils.py:36-39``)."""
+    if torch.cuda.is_available():
+        for t in tensors:
+            if t.is_cuda:
+                t.record_stream(torch.cuda.current_stream())
+    return tensors
+
+
+def _check_flash_arg(flash):
+    if flash not in ("cuda", "triton"):
+        raise ValueError(
+            "this MI355X-native BurstAttention has a single HIP tile backend "
+            "operating in the flash layout [B,S,N,H]; the reference's math-"
+            f"path layout (flash={flash!r}) is not supported"
+        )
+
+
+def burst_attn_func(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    softmax_scale: float = None,
+    flash: str = "cuda",
+    causal: bool = False,
+    optimize_bwd_comm: bool = False,
+    deterministic: bool = False,
+    process_group=None,
+    double_group=[None, None],
+):
+    return OpBurstAttn.apply(
+        q, k, v, softmax_scale, flash, causal, optimize_bwd_comm,
+        deterministic, process_group, double_group,
+    )
+
+
+def burst_attn_func_striped(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    softmax_scale: float = None,
+    flash: str = "cuda",
+    causal: bool = False,
+    optimize_bwd_comm: bool = False,
+    deterministic: bool = False,
+    process_group=None,
+    double_group=[None, None],
+):
+    return OpBurstAttnStrip.apply(
+        q, k, v, softmax_scale, flash, causal, optimize_bwd_comm,
+        deterministic, process_group, double_group,
+    )
+
+
+def _setup_ctx(ctx, q, softmax_scale, flash, causal, optimize_bwd_comm,
+               deterministic, process_group, double_group):
+    _check_flash_arg(flash)
+    if isinstance(double_group[0], tuple):
+        # [(group, dq_group), (group2, dq_group2)] — separate backward dq
+        # ring groups (reference :188-194)
+        ctx.dq_group = (double_group[0][1], double_group[1][1])
+        double_group = (double_group[0][0], double_group[1][0])
+    else:
+        ctx.dq_group = None
+    ctx.softmax_scale = (
+        1.0 / math.sqrt(q.shape[-1]) if softmax_scale is None else softmax_scale
+    )
+    ctx.causal = causal
+    ctx.optimize_bwd_comm = optimize_bwd_comm
+    ctx.deterministic = deterministic
+    ctx.process_group = process_group
+    ctx.double_group = double_group
+    return double_group
+
+
+def _finalize_fwd(ctx, q, ori_k, ori_v, o, lse):
+    out = o.to(dtype=q.dtype)
+    # lse [B,S,N,1] fp32 -> [B,N,S] contiguous (reference :250-252)
+    lse = lse.squeeze(-1).transpose(1, 2).contiguous()
+    ctx.save_for_backward(q, ori_k, ori_v, lse, replicate(out))
+    return out
+
+
+class OpBurstAttn(torch.autograd.Function):
+    """Zigzag (default) ring attention; q,k,v: [B, S/W, N, H]."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, softmax_scale=None, flash="cuda", causal=False,
+                optimize_bwd_comm=False, deterministic=False,
+                process_group=None, double_group=[None, None]):
+        double_group = _setup_ctx(
+            ctx, q, softmax_scale, flash, causal, optimize_bwd_comm,
+            deterministic, process_group, double_group,
+        )
+        P = get_tile_provider()
+        scale = ctx.softmax_scale
+        ring = Ring(process_group, double_group)
+        W, rank = ring.world_size, ring.rank
+        ori_k, ori_v = replicate(k), replicate(v)
+        comm_bufs = [torch.empty_like(k), torch.empty_like(v)]
+        half = q.shape[1] // 2
+        o = None   # fp32 [B,S,N,D]
+        lse = None  # fp32 [B,S,N,1]
+        for r in range(1, W + 1):
+            offset = get_partition_id(double_group, r)
+            split_kv = offset <= rank  # kv origin precedes this rank's chunks
+            if r != W:
+                ring.double_ring_send_recv([k, v], comm_bufs, r)
+                ring.commit()
+            if r == 1 or not causal:
+                o_i, lse_i = P.fwd(q, k, v, scale, causal)
+                if o is None:
+                    o = o_i.to(torch.float32)
+                    lse = lse_i.transpose(-2, -1).unsqueeze(-1).contiguous()
+                else:
+                    o, lse = P.merge(o, lse, o_i, lse_i)
+            elif split_kv:
+                # kv chunks [src, 2W-1-src] with src < rank: only chunk
+                # [src] (first half) is attended, by all local q (:225-231)
+                o_i, lse_i = P.fwd(q, k[:, :half], v[:, :half], scale, False)
+                o, lse = P.merge(o, lse, o_i, lse_i)
+            else:
+                # src > rank: only q's second half (chunk [2W-1-rank])
+                # attends, to the full received kv (:232-235)
+                o_i, lse_i = P.fwd(q[:, half:], k, v, scale, False)
+                o[:, half:], lse[:, half:] = P.merge(
+                    o[:, half:], lse[:, half:], o_i, lse_i
+                )
+            if r != W:
+                kv, comm_bufs = _record_stream(*comm_bufs), [k, v]
+                k, v = kv
+                ring.wait()
+        return _finalize_fwd(ctx, q, ori_k, ori_v, o, lse)
+
+    @staticmethod
+    def backward(ctx, grad_output):
+        q, k, v, lse, o = ctx.saved_tensors
+        P = get_tile_provider()
+        scale = ctx.softmax_scale
+        grad_output = grad_output.contiguous()
+        group, double_group = ctx.process_group, ctx.double_group
+        ring = Ring(group, double_group)
+        dq_ring = Ring(group, ctx.dq_group if ctx.dq_group is not None else double_group)
+        W, rank = ring.world_size, ring.rank
+        half = q.shape[1] // 2
+
+        dq = torch.zeros(q.shape, dtype=torch.float32, device=q.device)
+        dk = torch.zeros(k.shape, dtype=torch.float32, device=k.device)
+        dv = torch.zeros(v.shape, dtype=torch.float32, device=v.device)
+
+        if ctx.optimize_bwd_comm:
+            # ring the tiny fp32 delta instead of o (reference :269-278)
+            dlt = P.bwd_preprocess(o, grad_output)  # [B,N,S] fp32
+        else:
+            dlt = o  # o travels; delta recomputed per round
+
+        read_bufs = [torch.empty_like(t) for t in (dlt, grad_output, q, lse)]
+        dq_buf = [torch.empty_like(dq)]
+        for r in range(1, W + 1):
+            offset = get_partition_id(double_group, r)
+            split_q = offset <= rank  # q origin precedes this rank
+            if r != W:
+                ring.double_ring_send_recv([dlt, grad_output, q, lse], read_bufs, r)
+                ring.commit()
+            if r != 1:
+                dq_ring.double_ring_send_recv_q([dq], dq_buf, r)
+                dq_ring.commit()
+            delta = dlt if ctx.optimize_bwd_comm else P.bwd_preprocess(dlt, grad_output)
+            if r == 1 or not ctx.causal:
+                dq_i, dk_i, dv_i = P.bwd(
+                    grad_output, q, k, v, delta, lse, scale, ctx.causal,
+                    ctx.deterministic,
+                )
+                acc = "full"
+            elif split_q:
+                # travelling q's second half attends my full kv (:322-345)
+                dq_i, dk_i, dv_i = P.bwd(
+                    grad_output[:, half:], q[:, half:], k, v,
+                    delta[:, :, half:], lse[:, :, half:], scale, False,
+                    ctx.deterministic,
+                )
+                acc = "q_half"
+            else:
+                # travelling q (all of it) attends only my kv first half
+                # (:347-367)
+                dq_i, dk_i, dv_i = P.bwd(
+                    grad_output, q, k[:, :half], v[:, :half], delta, lse,
+                    scale, False, ctx.deterministic,
+                )
+                acc = "kv_half"
+            if r != W:
+                recv, read_bufs = (
+                    _record_stream(*read_bufs),
+                    [dlt, grad_output, q, lse],
+                )
+                dlt, grad_output, q, lse = recv
+            ring.wait()
+            if r != 1:
+                dq_ring.wait()
+                recv, dq_buf = _record_stream(*dq_buf), [dq]
+                dq = recv[0]
+            if acc == "full":
+                dq += dq_i
+                dk += dk_i
+                dv += dv_i
+            elif acc == "q_half":
+                dq[:, half:] += dq_i
+                dk += dk_i
+                dv += dv_i
+            else:
+                dq += dq_i
+                dk[:, :half] += dk_i
+                dv[:, :half] += dv_i
+        # one extra hop returns the travelling dq to its owner (:393-396)
+        dq_ring.double_ring_send_recv_q([dq], dq_buf, W + 1)
+        dq_ring.commit()
+        dq_ring.wait()
+        dq = _record_stream(*dq_buf)[0]
+        return (
+            dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype),
+            None, None, None, None, None, None, None,
+        )
+
+
+class OpBurstAttnStrip(torch.autograd.Function):
+    """Striped ring attention (token t → rank t mod W); q,k,v [B,S/W,N,H]."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, softmax_scale=None, flash="cuda", causal=False,
+                optimize_bwd_comm=False, deterministic=False,
+                process_group=None, double_group=[None, None]):
+        double_group = _setup_ctx(
+            ctx, q, softmax_scale, flash, causal, optimize_bwd_comm,
+            deterministic, process_group, double_group,
+        )
+        P = get_tile_provider()
+        scale = ctx.softmax_scale
+        ring = Ring(process_group, double_group)
+        W, rank = ring.world_size, ring.rank
+        ori_k, ori_v = replicate(k), replicate(v)
+        comm_bufs = [torch.empty_like(k), torch.empty_like(v)]
+        o = None
+        lse = None
+        for r in range(1, W + 1):
+            offset = get_partition_id(double_group, r)
+            causal_shift = offset > rank  # kv origin follows this rank
+            if r != W:
+                ring.double_ring_send_recv([k, v], comm_bufs, r)
+                ring.commit()
+            if not causal_shift or not causal:
+                o_i, lse_i = P.fwd(q, k, v, scale, causal)
+                if o is None:
+                    o = o_i.to(torch.float32)
+                    lse = lse_i.transpose(-2, -1).unsqueeze(-1).contiguous()
+                else:
+                    o, lse = P.merge(o, lse, o_i, lse_i)
+            else:
+                # one-token shift: striped causal vs a later rank's kv
+                # (:463-475)
+                o_i, lse_i = P.fwd(q[:, 1:], k[:, :-1], v[:, :-1], scale, causal)
+                o[:, 1:], lse[:, 1:] = P.merge(o[:, 1:], lse[:, 1:], o_i, lse_i)
+            if r != W:
+                kv, comm_bufs = _record_stream(*comm_bufs), [k, v]
+                k, v = kv
+                ring.wait()
+        return _finalize_fwd(ctx, q, ori_k, ori_v, o, lse)
+
+    @staticmethod
+    def backward(ctx, grad_output):
+        q, k, v, lse, o = ctx.saved_tensors
+        P = get_tile_provider()
+        scale = ctx.softmax_scale
+        grad_output = grad_output.contiguous()
+        group, double_group = ctx.process_group, ctx.double_group
+        ring = Ring(group, double_group)
+        dq_ring = Ring(group, ctx.dq_group if ctx.dq_group is not None else double_group)
+        W, rank = ring.world_size, ring.rank
+
+        dq = torch.zeros(q.shape, dtype=torch.float32, device=q.device)
+        dk = torch.zeros(k.shape, dtype=torch.float32, device=k.device)
+        dv = torch.zeros(v.shape, dtype=torch.float32, device=v.device)
+
+        if ctx.optimize_bwd_comm:
+            dlt = P.bwd_preprocess(o, grad_output)
+        else:
+            dlt = o
+
+        read_bufs = [torch.empty_like(t) for t in (dlt, grad_output, q, lse)]
+        dq_buf = [torch.empty_like(dq)]
+        for r in range(1, W + 1):
+            offset = get_partition_id(double_group, r)
+            causal_shift = offset <= rank and r != 1  # q origin precedes
+            if r != W:
+                ring.double_ring_send_recv([dlt, grad_output, q, lse], read_bufs, r)
+                ring.commit()
+            if r != 1:
+                dq_ring.double_ring_send_recv_q([dq], dq_buf, r)
+                dq_ring.commit()
+            delta = dlt if ctx.optimize_bwd_comm else P.bwd_preprocess(dlt, grad_output)
+            if not causal_shift or not ctx.causal:
+                dq_i, dk_i, dv_i = P.bwd(
+                    grad_output, q, k, v, delta, lse, scale, ctx.causal,
+                    ctx.deterministic,
+                )
+                shifted = False
+            else:
+                # shifted tile: q[1:] vs k[:-1] (reference :557-585)
+                dq_i, dk_i, dv_i = P.bwd(
+                    grad_output[:, 1:], q[:, 1:], k[:, :-1], v[:, :-1],
+                    delta[:, :, 1:], lse[:, :, 1:], scale, ctx.causal,
+                    ctx.deterministic,
+                )
+                shifted = True
+            if r != W:
+                recv, read_bufs = (
+                    _record_stream(*read_bufs),
+                    [dlt, grad_output, q, lse],
+                )
+                dlt, grad_output, q, lse = recv
+            ring.wait()
+            if r != 1:
+                dq_ring.wait()
+                recv, dq_buf = _record_stream(*dq_buf), [dq]
+                dq = recv[0]
+            if not shifted:
+                dq += dq_i
+                dk += dk_i
+                dv += dv_i
+            else:
+                dq[:, 1:] += dq_i
+                dk[:, :-1] += dk_i
+                dv[:, :-1] += dv_i
+        dq_ring.double_ring_send_recv_q([dq], dq_buf, W + 1)
+        dq_ring.commit()
+        dq_ring.wait()
+        dq = _record_stream(*dq_buf)[0]
+        return (
+            dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype),
+            None, None, None, None, None, None, None,
+        )
