@@ -1,0 +1,90 @@
+"""In-tree build of the gfx950 HIP extension (burst_attn_amd/_C.so).
+
+Driven by hipcc directly (no JIT cache — the .so must live in-tree so the
+gpurun snapshot carries it).  Called by __graft_entry__.build().
+"""
+
+import os
+import subprocess
+import sys
+import sysconfig
+
+PKG_DIR = os.path.dirname(os.path.abspath(__file__))
+CSRC = os.path.join(PKG_DIR, "csrc")
+BUILD = os.path.join(PKG_DIR, "csrc", "_build")
+ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
+HIPCC = os.environ.get("HIPCC", "/opt/rocm/bin/hipcc")
+
+DEVICE_SOURCES = ["attn_fwd.hip", "attn_bwd.hip"]
+HOST_SOURCES = ["ext_torch.cpp"]
+
+
+def _run(cmd, verbose):
+    if verbose:
+        print("+ " + " ".join(cmd), flush=True)
+    r = subprocess.run(cmd, capture_output=True, text=True)
+    if r.returncode != 0:
+        sys.stderr.write(r.stdout + r.stderr)
+        raise RuntimeError(f"build command failed: {' '.join(cmd[:3])} ...")
+    return r
+
+
+def _newer(src, dst):
+    return not os.path.exists(dst) or os.path.getmtime(src) > os.path.getmtime(dst)
+
+
+def build(verbose=True, force=False):
+    import torch
+    from torch.utils import cpp_extension
+
+    os.makedirs(BUILD, exist_ok=True)
+    out_so = os.path.join(PKG_DIR, "_C.so")
+
+    torch_includes = cpp_extension.include_paths()
+    py_include = sysconfig.get_paths()["include"]
+    abi = int(torch._C._GLIBCXX_USE_CXX11_ABI)
+
+    common = [
+        f"--offload-arch={ARCH}", "-O3", "-std=c++17", "-fPIC",
+        "-DNDEBUG",
+    ]
+    objs = []
+    hdr = os.path.join(CSRC, "attn_common.h")
+    abihdr = os.path.join(PKG_DIR, "..", "include", "burst_attn_hip.h")
+    for src in DEVICE_SOURCES:
+        sp = os.path.join(CSRC, src)
+        op = os.path.join(BUILD, src.replace(".hip", ".o"))
+        objs.append(op)
+        if force or _newer(sp, op) or _newer(hdr, op) or _newer(abihdr, op):
+            _run([HIPCC, *common, "-c", sp, "-o", op], verbose)
+    for src in HOST_SOURCES:
+        sp = os.path.join(CSRC, src)
+        op = os.path.join(BUILD, src.replace(".cpp", ".o"))
+        objs.append(op)
+        if force or _newer(sp, op) or _newer(abihdr, op):
+            cmd = [
+                HIPCC, *common,
+                f"-D_GLIBCXX_USE_CXX11_ABI={abi}",
+                "-DTORCH_EXTENSION_NAME=_C",
+                "-DTORCH_API_INCLUDE_EXTENSION_H",
+                "-DUSE_ROCM=1", "-D__HIP_PLATFORM_AMD__=1",
+            ]
+            for inc in torch_includes + [py_include]:
+                cmd += ["-I", inc]
+            cmd += ["-c", sp, "-o", op]
+            _run(cmd, verbose)
+    if force or any(_newer(o, out_so) for o in objs):
+        lib_dirs = cpp_extension.library_paths()
+        cmd = [HIPCC, "-shared", "-fPIC", *objs, "-o", out_so]
+        for ld in lib_dirs:
+            cmd += [f"-L{ld}", f"-Wl,-rpath,{ld}"]
+        cmd += ["-ltorch", "-ltorch_cpu", "-ltorch_python", "-lc10",
+                "-ltorch_hip", "-lc10_hip", "-lamdhip64"]
+        _run(cmd, verbose)
+    if verbose:
+        print(f"built {out_so}")
+    return out_so
+
+
+if __name__ == "__main__":
+    build(force="--force" in sys.argv)

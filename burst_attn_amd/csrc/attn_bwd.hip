@@ -1,0 +1,483 @@
+// BurstAttention backward tile kernels for gfx950 (MI355X, CDNA4).
+//
+// Replaces the flash-attn backward the reference calls at
+// burst_attn/burst_utils.py:211-248, with the tile math of the reference's
+// own restatement (burst_utils.py:77-101):
+//     p   = exp(q k^T * scale - lse)
+//     dv += p^T do;   dp = do v^T;   ds = p * (dp - delta) * scale
+//     dq += ds k;     dk += ds^T q
+//
+// Split into three kernels (two-pass dq: deterministic by construction, no
+// atomics — the `deterministic` flag is accepted and always honoured):
+//   1. bwd_preprocess: delta = rowsum(o * do) fp32   (flash's preprocess)
+//   2. dq kernel  (8 waves, q-block resident, streams k/v):
+//        S^T = mfma(K,Q); dP^T = mfma(V,dO); dS^T in-lane;
+//        dQ^T += mfma(K^T, dS^T)            — all softmax state lane-local
+//   3. dkdv kernel (4 waves, kv-block resident, streams q/do):
+//        S = mfma(Q,K^T); dP = mfma(dO,V^T); dS in-lane;
+//        dV^T += mfma(dO^T, P); dK^T += mfma(Q^T, dS)
+// Operand scheme and LDS swizzle: see attn_common.h.
+
+#include "attn_common.h"
+#include "../../include/burst_attn_hip.h"
+
+#include <stdio.h>
+
+namespace {
+
+// ====================== delta preprocess ==============================
+template <typename T, int D>
+__global__ __launch_bounds__(256) void bwd_preprocess_kernel(
+    const T* __restrict__ o, const T* __restrict__ dout,
+    float* __restrict__ delta, int S, int N,
+    int64_t o_sb, int64_t o_ss, int64_t o_sh,
+    int64_t g_sb, int64_t g_ss, int64_t g_sh) {
+  // one wave: 8 rows x 8 lanes/row, 16B..32B per lane
+  constexpr int EPL = D / 8;  // elements per lane (16 for D=128)
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int row_in = wave * 8 + (lane >> 3);
+  const int sub = lane & 7;
+  const int s = blockIdx.x * 32 + row_in;
+  const int n = blockIdx.y, b = blockIdx.z;
+  if (s >= S) return;
+  const T* op = o + b * o_sb + (int64_t)s * o_ss + n * o_sh + sub * EPL;
+  const T* gp = dout + b * g_sb + (int64_t)s * g_ss + n * g_sh + sub * EPL;
+  float acc = 0.f;
+#pragma unroll
+  for (int j = 0; j < EPL; ++j) acc += (float)op[j] * (float)gp[j];
+  acc += __shfl_xor(acc, 1);
+  acc += __shfl_xor(acc, 2);
+  acc += __shfl_xor(acc, 4);
+  if (sub == 0) delta[((int64_t)b * N + n) * S + s] = acc;
+}
+
+// ====================== dq kernel (q-resident) ========================
+constexpr int KVBLK = 64;
+
+template <typename T, int D>
+__global__ __launch_bounds__(512) void bwd_dq_kernel(
+    const T* __restrict__ dout, const T* __restrict__ q,
+    const T* __restrict__ k, const T* __restrict__ v,
+    const float* __restrict__ delta, const float* __restrict__ lse,
+    float* __restrict__ dq, int Sq, int Sk, int N,
+    int64_t g_sb, int64_t g_ss, int64_t g_sh,
+    int64_t q_sb, int64_t q_ss, int64_t q_sh,
+    int64_t k_sb, int64_t k_ss, int64_t k_sh,
+    int64_t v_sb, int64_t v_ss, int64_t v_sh,
+    int64_t d_sb, int64_t d_sh, int64_t l_sb, int64_t l_sh,
+    float scale, int causal) {
+  using MT = mfma_traits<T>;
+  using frag = typename MT::frag;
+  constexpr int SWZ = (D == 128) ? 15 : 7;   // row-major images (2*D B rows)
+  constexpr int SWZ_T = 7;                   // transposed images (128 B rows)
+  constexpr int NT = 512;
+  constexpr int PT = (KVBLK * D / 8) / NT;
+
+  // [2 buffers][K row-major | V row-major | K^T transposed]
+  __shared__ T lds[2 * 3 * KVBLK * D];
+  auto ldsK = [&](int buf) -> T* { return lds + buf * (3 * KVBLK * D); };
+  auto ldsV = [&](int buf) -> T* { return lds + buf * (3 * KVBLK * D) + KVBLK * D; };
+  auto ldsKT = [&](int buf) -> T* { return lds + buf * (3 * KVBLK * D) + 2 * KVBLK * D; };
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int l31 = lane & 31, hi = lane >> 5;
+  const int n = blockIdx.y, b = blockIdx.z;
+  const int qb = blockIdx.x * 256 + wave * 32;
+  const int q_row = qb + l31;
+
+  const T* qp = q + b * q_sb + (int64_t)n * q_sh;
+  const T* gp = dout + b * g_sb + (int64_t)n * g_sh;
+  const T* kp = k + b * k_sb + (int64_t)n * k_sh;
+  const T* vp = v + b * v_sb + (int64_t)n * v_sh;
+
+  frag qf[D / 16], gf[D / 16];
+#pragma unroll
+  for (int s = 0; s < D / 16; ++s) {
+    if (q_row < Sq) {
+      qf[s] = __builtin_bit_cast(
+          frag, *(const u32x4_t*)(qp + (int64_t)q_row * q_ss + 16 * s + 8 * hi));
+      gf[s] = __builtin_bit_cast(
+          frag, *(const u32x4_t*)(gp + (int64_t)q_row * g_ss + 16 * s + 8 * hi));
+    } else {
+      u32x4_t z = {0, 0, 0, 0};
+      qf[s] = __builtin_bit_cast(frag, z);
+      gf[s] = __builtin_bit_cast(frag, z);
+    }
+  }
+  const float c2 = scale * BA_LOG2E;
+  const float lse2 =
+      (q_row < Sq) ? lse[b * l_sb + n * l_sh + q_row] * BA_LOG2E : 0.f;
+  const float dlt =
+      (q_row < Sq) ? delta[b * d_sb + n * d_sh + q_row] : 0.f;
+
+  f32x16_t dqt[D / 32];
+#pragma unroll
+  for (int dt = 0; dt < D / 32; ++dt) dqt[dt] = (f32x16_t)(0.f);
+
+  const int kv_limit = causal ? min(Sk, (int)(blockIdx.x + 1) * 256) : Sk;
+  const int nt = (kv_limit + KVBLK - 1) / KVBLK;
+
+  auto issue_loads = [&](int tile, u32x4_t* kreg, u32x4_t* vreg) {
+    const int kv0 = tile * KVBLK;
+#pragma unroll
+    for (int c = 0; c < PT; ++c) {
+      const int flat = tid + c * NT;
+      const int row = flat / (D / 8), col8 = flat % (D / 8);
+      const int kvg = kv0 + row;
+      if (kvg < Sk) {
+        kreg[c] = *(const u32x4_t*)(kp + (int64_t)kvg * k_ss + col8 * 8);
+        vreg[c] = *(const u32x4_t*)(vp + (int64_t)kvg * v_ss + col8 * 8);
+      } else {
+        u32x4_t z = {0, 0, 0, 0};
+        kreg[c] = z;
+        vreg[c] = z;
+      }
+    }
+  };
+  auto write_lds = [&](int buf, const u32x4_t* kreg, const u32x4_t* vreg) {
+#pragma unroll
+    for (int c = 0; c < PT; ++c) {
+      const int flat = tid + c * NT;
+      const int row = flat / (D / 8), col8 = flat % (D / 8);
+      const int byte = ba_swz<SWZ>(row * (2 * D) + col8 * 16, row);
+      *(u32x4_t*)((char*)ldsK(buf) + byte) = kreg[c];
+      *(u32x4_t*)((char*)ldsV(buf) + byte) = vreg[c];
+      ba_st_transposed<T, KVBLK, SWZ_T>(ldsKT(buf), row, col8 * 8, kreg[c]);
+    }
+  };
+
+  {
+    u32x4_t kreg[PT], vreg[PT];
+    issue_loads(0, kreg, vreg);
+    write_lds(0, kreg, vreg);
+    __syncthreads();
+  }
+
+  int cur = 0;
+  for (int t = 0; t < nt; ++t) {
+    const int kv0 = t * KVBLK;
+    const bool has_next = (t + 1) < nt;
+    u32x4_t kreg[PT], vreg[PT];
+    if (has_next) issue_loads(t + 1, kreg, vreg);
+
+    const bool active = !causal || (kv0 <= qb + 31);
+    if (active) {
+#pragma unroll
+      for (int kvs = 0; kvs < 2; ++kvs) {
+        f32x16_t st = (f32x16_t)(0.f), dpt = (f32x16_t)(0.f);
+#pragma unroll
+        for (int s = 0; s < D / 16; ++s) {
+          frag kf = ba_ld_rowslice<T, D, SWZ>(ldsK(cur), kvs * 32 + l31,
+                                              16 * s + 8 * hi);
+          frag vf = ba_ld_rowslice<T, D, SWZ>(ldsV(cur), kvs * 32 + l31,
+                                              16 * s + 8 * hi);
+          st = MT::mma(kf, qf[s], st);
+          dpt = MT::mma(vf, gf[s], dpt);
+        }
+        // p^T = exp2(S^T*c2 - lse2);  dS^T = p^T*(dP^T - delta)*scale
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int kv_g = kv0 + kvs * 32 + ba_crow(r, 0) + 4 * hi;
+          const bool valid =
+              kv_g < Sk && (!causal || kv_g <= q_row) && q_row < Sq;
+          const float e = valid ? st[r] * c2 - lse2 : BA_NEG_BIG;
+          const float p = exp2f(e);
+          st[r] = p * (dpt[r] - dlt) * scale;
+        }
+        frag dsf[2];
+        ba_build_frag_pair<T>(st, dsf);
+#pragma unroll
+        for (int dt = 0; dt < D / 32; ++dt) {
+          const int col = dt * 32 + l31;
+#pragma unroll
+          for (int u = 0; u < 2; ++u) {
+            frag ktf = ba_ld_rowslice<T, KVBLK, SWZ_T>(
+                ldsKT(cur), col, kvs * 32 + 16 * u + 8 * hi);
+            dqt[dt] = MT::mma(ktf, dsf[u], dqt[dt]);
+          }
+        }
+      }
+    }
+    if (has_next) write_lds(cur ^ 1, kreg, vreg);
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  if (q_row < Sq) {
+    float* drow = dq + (((int64_t)b * Sq + q_row) * N + n) * D;
+#pragma unroll
+    for (int dt = 0; dt < D / 32; ++dt)
+#pragma unroll
+      for (int r = 0; r < 16; ++r)
+        drow[dt * 32 + ba_crow(r, hi)] = dqt[dt][r];
+  }
+}
+
+// ====================== dk/dv kernel (kv-resident) ====================
+constexpr int QBLK = 64;
+
+template <typename T, int D>
+__global__ __launch_bounds__(256) void bwd_dkdv_kernel(
+    const T* __restrict__ dout, const T* __restrict__ q,
+    const T* __restrict__ k, const T* __restrict__ v,
+    const float* __restrict__ delta, const float* __restrict__ lse,
+    float* __restrict__ dk, float* __restrict__ dv, int Sq, int Sk, int N,
+    int64_t g_sb, int64_t g_ss, int64_t g_sh,
+    int64_t q_sb, int64_t q_ss, int64_t q_sh,
+    int64_t k_sb, int64_t k_ss, int64_t k_sh,
+    int64_t v_sb, int64_t v_ss, int64_t v_sh,
+    int64_t d_sb, int64_t d_sh, int64_t l_sb, int64_t l_sh,
+    float scale, int causal) {
+  using MT = mfma_traits<T>;
+  using frag = typename MT::frag;
+  constexpr int SWZ = (D == 128) ? 15 : 7;
+  constexpr int SWZ_T = 7;
+  constexpr int NT = 256;  // 4 waves -> 1 wave/SIMD, accs live in AGPRs
+  constexpr int PT = (QBLK * D / 8) / NT;
+
+  // [2 buffers][Q | dO row-major | Q^T | dO^T transposed]
+  __shared__ T lds[2 * 4 * QBLK * D];
+  auto ldsQ = [&](int buf) -> T* { return lds + buf * (4 * QBLK * D); };
+  auto ldsG = [&](int buf) -> T* { return lds + buf * (4 * QBLK * D) + QBLK * D; };
+  auto ldsQT = [&](int buf) -> T* { return lds + buf * (4 * QBLK * D) + 2 * QBLK * D; };
+  auto ldsGT = [&](int buf) -> T* { return lds + buf * (4 * QBLK * D) + 3 * QBLK * D; };
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int l31 = lane & 31, hi = lane >> 5;
+  const int n = blockIdx.y, b = blockIdx.z;
+  const int kvb = blockIdx.x * 128 + wave * 32;
+  const int kv_col = kvb + l31;
+
+  const T* qp = q + b * q_sb + (int64_t)n * q_sh;
+  const T* gp = dout + b * g_sb + (int64_t)n * g_sh;
+  const T* kp = k + b * k_sb + (int64_t)n * k_sh;
+  const T* vp = v + b * v_sb + (int64_t)n * v_sh;
+  const float* dp_ = delta + b * d_sb + n * d_sh;
+  const float* lp_ = lse + b * l_sb + n * l_sh;
+
+  // resident K/V fragments: B-operand (col = kv) of S = mfma(Q, K^T)
+  frag kf[D / 16], vf[D / 16];
+#pragma unroll
+  for (int s = 0; s < D / 16; ++s) {
+    if (kv_col < Sk) {
+      kf[s] = __builtin_bit_cast(
+          frag, *(const u32x4_t*)(kp + (int64_t)kv_col * k_ss + 16 * s + 8 * hi));
+      vf[s] = __builtin_bit_cast(
+          frag, *(const u32x4_t*)(vp + (int64_t)kv_col * v_ss + 16 * s + 8 * hi));
+    } else {
+      u32x4_t z = {0, 0, 0, 0};
+      kf[s] = __builtin_bit_cast(frag, z);
+      vf[s] = __builtin_bit_cast(frag, z);
+    }
+  }
+  const float c2 = scale * BA_LOG2E;
+
+  f32x16_t dvt[D / 32], dkt[D / 32];
+#pragma unroll
+  for (int dt = 0; dt < D / 32; ++dt) {
+    dvt[dt] = (f32x16_t)(0.f);
+    dkt[dt] = (f32x16_t)(0.f);
+  }
+
+  // causal: q tiles below this block's first kv row are fully masked
+  const int t0 = causal ? (blockIdx.x * 128) / QBLK : 0;
+  const int nt = (Sq + QBLK - 1) / QBLK;
+
+  auto issue_loads = [&](int tile, u32x4_t* qreg, u32x4_t* greg) {
+    const int q0 = tile * QBLK;
+#pragma unroll
+    for (int c = 0; c < PT; ++c) {
+      const int flat = tid + c * NT;
+      const int row = flat / (D / 8), col8 = flat % (D / 8);
+      const int qg = q0 + row;
+      if (qg < Sq) {
+        qreg[c] = *(const u32x4_t*)(qp + (int64_t)qg * q_ss + col8 * 8);
+        greg[c] = *(const u32x4_t*)(gp + (int64_t)qg * g_ss + col8 * 8);
+      } else {
+        u32x4_t z = {0, 0, 0, 0};
+        qreg[c] = z;
+        greg[c] = z;
+      }
+    }
+  };
+  auto write_lds = [&](int buf, const u32x4_t* qreg, const u32x4_t* greg) {
+#pragma unroll
+    for (int c = 0; c < PT; ++c) {
+      const int flat = tid + c * NT;
+      const int row = flat / (D / 8), col8 = flat % (D / 8);
+      const int byte = ba_swz<SWZ>(row * (2 * D) + col8 * 16, row);
+      *(u32x4_t*)((char*)ldsQ(buf) + byte) = qreg[c];
+      *(u32x4_t*)((char*)ldsG(buf) + byte) = greg[c];
+      ba_st_transposed<T, QBLK, SWZ_T>(ldsQT(buf), row, col8 * 8, qreg[c]);
+      ba_st_transposed<T, QBLK, SWZ_T>(ldsGT(buf), row, col8 * 8, greg[c]);
+    }
+  };
+
+  {
+    u32x4_t qreg[PT], greg[PT];
+    issue_loads(t0, qreg, greg);
+    write_lds(0, qreg, greg);
+    __syncthreads();
+  }
+
+  int cur = 0;
+  for (int t = t0; t < nt; ++t) {
+    const int q0 = t * QBLK;
+    const bool has_next = (t + 1) < nt;
+    u32x4_t qreg[PT], greg[PT];
+    if (has_next) issue_loads(t + 1, qreg, greg);
+
+    // this wave's kv rows need q >= kvb to have any unmasked element
+    const bool active = !causal || (q0 + QBLK - 1 >= kvb);
+    if (active) {
+#pragma unroll
+      for (int qs = 0; qs < 2; ++qs) {
+        f32x16_t st = (f32x16_t)(0.f), dpt = (f32x16_t)(0.f);
+#pragma unroll
+        for (int s = 0; s < D / 16; ++s) {
+          frag qfr = ba_ld_rowslice<T, D, SWZ>(ldsQ(cur), qs * 32 + l31,
+                                               16 * s + 8 * hi);
+          frag gfr = ba_ld_rowslice<T, D, SWZ>(ldsG(cur), qs * 32 + l31,
+                                               16 * s + 8 * hi);
+          st = MT::mma(qfr, kf[s], st);
+          dpt = MT::mma(gfr, vf[s], dpt);
+        }
+        f32x16_t ds;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int q_g = q0 + qs * 32 + ba_crow(r, 0) + 4 * hi;
+          const bool valid =
+              q_g < Sq && kv_col < Sk && (!causal || q_g >= kv_col);
+          // per-q-row lse/delta: broadcast loads (same addr in lane group)
+          const float l2 = valid ? lp_[q_g] * BA_LOG2E : 0.f;
+          const float dl = valid ? dp_[q_g] : 0.f;
+          const float e = valid ? st[r] * c2 - l2 : BA_NEG_BIG;
+          const float p = exp2f(e);
+          st[r] = p;                             // P
+          ds[r] = p * (dpt[r] - dl) * scale;     // dS
+        }
+        frag pf[2], dsf[2];
+        ba_build_frag_pair<T>(st, pf);
+        ba_build_frag_pair<T>(ds, dsf);
+#pragma unroll
+        for (int dt = 0; dt < D / 32; ++dt) {
+          const int col = dt * 32 + l31;
+#pragma unroll
+          for (int u = 0; u < 2; ++u) {
+            frag gtf = ba_ld_rowslice<T, QBLK, SWZ_T>(
+                ldsGT(cur), col, qs * 32 + 16 * u + 8 * hi);
+            dvt[dt] = MT::mma(gtf, pf[u], dvt[dt]);
+            frag qtf = ba_ld_rowslice<T, QBLK, SWZ_T>(
+                ldsQT(cur), col, qs * 32 + 16 * u + 8 * hi);
+            dkt[dt] = MT::mma(qtf, dsf[u], dkt[dt]);
+          }
+        }
+      }
+    }
+    if (has_next) write_lds(cur ^ 1, qreg, greg);
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  if (kv_col < Sk) {
+    float* dvrow = dv + (((int64_t)b * Sk + kv_col) * N + n) * D;
+    float* dkrow = dk + (((int64_t)b * Sk + kv_col) * N + n) * D;
+#pragma unroll
+    for (int dt = 0; dt < D / 32; ++dt)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        dvrow[dt * 32 + ba_crow(r, hi)] = dvt[dt][r];
+        dkrow[dt * 32 + ba_crow(r, hi)] = dkt[dt][r];
+      }
+  }
+}
+
+}  // namespace
+
+// ====================== launchers =====================================
+extern "C" int bahip_attn_bwd_preprocess(
+    const void* o, const void* dout, float* delta, int64_t B, int64_t S,
+    int64_t N, int64_t D, const int64_t o_strides[3],
+    const int64_t do_strides[3], int o_dtype, int do_dtype, void* stream) {
+  if (o_dtype != do_dtype) return 1003;
+  dim3 grid((unsigned)((S + 31) / 32), (unsigned)N, (unsigned)B);
+#define LAUNCH_PRE(T, DD)                                                   \
+  bwd_preprocess_kernel<T, DD><<<grid, 256, 0, (hipStream_t)stream>>>(      \
+      (const T*)o, (const T*)dout, delta, (int)S, (int)N, o_strides[0],     \
+      o_strides[1], o_strides[2], do_strides[0], do_strides[1],             \
+      do_strides[2])
+  if (D == 128 && o_dtype == BAHIP_BF16) LAUNCH_PRE(__bf16, 128);
+  else if (D == 128 && o_dtype == BAHIP_F16) LAUNCH_PRE(_Float16, 128);
+  else if (D == 64 && o_dtype == BAHIP_BF16) LAUNCH_PRE(__bf16, 64);
+  else if (D == 64 && o_dtype == BAHIP_F16) LAUNCH_PRE(_Float16, 64);
+  else return 1002;
+#undef LAUNCH_PRE
+  BA_CHECK_LAUNCH();
+  return 0;
+}
+
+template <typename T, int D>
+static int launch_bwd(const void* dout, const void* q, const void* k,
+                      const void* v, const float* delta, const float* lse,
+                      float* dq, float* dk, float* dv, int64_t B, int64_t Sq,
+                      int64_t Sk, int64_t N, const int64_t* gs,
+                      const int64_t* qs, const int64_t* ks, const int64_t* vs,
+                      const int64_t* ds, const int64_t* ls, float scale,
+                      int causal, void* stream) {
+  dim3 grid_dq((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B);
+  bwd_dq_kernel<T, D><<<grid_dq, 512, 0, (hipStream_t)stream>>>(
+      (const T*)dout, (const T*)q, (const T*)k, (const T*)v, delta, lse, dq,
+      (int)Sq, (int)Sk, (int)N, gs[0], gs[1], gs[2], qs[0], qs[1], qs[2],
+      ks[0], ks[1], ks[2], vs[0], vs[1], vs[2], ds[0], ds[1], ls[0], ls[1],
+      scale, causal);
+  BA_CHECK_LAUNCH();
+  dim3 grid_kv((unsigned)((Sk + 127) / 128), (unsigned)N, (unsigned)B);
+  bwd_dkdv_kernel<T, D><<<grid_kv, 256, 0, (hipStream_t)stream>>>(
+      (const T*)dout, (const T*)q, (const T*)k, (const T*)v, delta, lse, dk,
+      dv, (int)Sq, (int)Sk, (int)N, gs[0], gs[1], gs[2], qs[0], qs[1], qs[2],
+      ks[0], ks[1], ks[2], vs[0], vs[1], vs[2], ds[0], ds[1], ls[0], ls[1],
+      scale, causal);
+  BA_CHECK_LAUNCH();
+  return 0;
+}
+
+extern "C" int bahip_attn_bwd(
+    const void* dout, const void* q, const void* k, const void* v,
+    const float* delta, const float* lse, float* dq, float* dk, float* dv,
+    int64_t B, int64_t Sq, int64_t Sk, int64_t N, int64_t D,
+    const int64_t do_strides[3], const int64_t q_strides[3],
+    const int64_t k_strides[3], const int64_t v_strides[3],
+    const int64_t delta_strides[2], const int64_t lse_strides[2],
+    float softmax_scale, int causal, int deterministic, int dtype,
+    void* stream) {
+  (void)deterministic;  // two-pass dq is deterministic by construction
+  if (causal && Sq != Sk) return 1001;
+  if (D == 128 && dtype == BAHIP_BF16)
+    return launch_bwd<__bf16, 128>(dout, q, k, v, delta, lse, dq, dk, dv, B,
+                                   Sq, Sk, N, do_strides, q_strides, k_strides,
+                                   v_strides, delta_strides, lse_strides,
+                                   softmax_scale, causal, stream);
+  if (D == 128 && dtype == BAHIP_F16)
+    return launch_bwd<_Float16, 128>(dout, q, k, v, delta, lse, dq, dk, dv, B,
+                                     Sq, Sk, N, do_strides, q_strides,
+                                     k_strides, v_strides, delta_strides,
+                                     lse_strides, softmax_scale, causal,
+                                     stream);
+  if (D == 64 && dtype == BAHIP_BF16)
+    return launch_bwd<__bf16, 64>(dout, q, k, v, delta, lse, dq, dk, dv, B, Sq,
+                                  Sk, N, do_strides, q_strides, k_strides,
+                                  v_strides, delta_strides, lse_strides,
+                                  softmax_scale, causal, stream);
+  if (D == 64 && dtype == BAHIP_F16)
+    return launch_bwd<_Float16, 64>(dout, q, k, v, delta, lse, dq, dk, dv, B,
+                                    Sq, Sk, N, do_strides, q_strides,
+                                    k_strides, v_strides, delta_strides,
+                                    lse_strides, softmax_scale, causal,
+                                    stream);
+  return 1002;
+}

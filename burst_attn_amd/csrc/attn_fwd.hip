@@ -1,0 +1,307 @@
+// BurstAttention forward tile kernel for gfx950 (MI355X, CDNA4).
+//
+// Computes one flash-attention tile o = softmax(q k^T * scale) v plus its
+// log-sum-exp — the role the flash-attn CUDA extension plays for the
+// reference at burst_attn/burst_utils.py:149-177 — as a single hand-written
+// HIP kernel (online softmax in the exp2 domain, fused, no S matrix in HBM).
+//
+// Structure (see attn_common.h header comment for the operand scheme):
+//   * workgroup = 8 waves (512 threads); wave w owns q rows
+//     [blk*256 + w*32, +32); grid = (ceil(Sq/256), N, B).
+//   * kv tiles of KVBLK=64 rows staged in LDS, double-buffered, one barrier
+//     per tile; next tile's global loads are issued before the current
+//     tile's MFMAs (async-stage split).  K is stored row-major [kv][D]
+//     (XOR-swizzled); V is stored TRANSPOSED [D][kv] at staging time so
+//     that both operands' MFMA fragments are plain 16B row-slice
+//     ds_read_b128s — no per-element transposed gathers (those made the
+//     compiler hoist & spill hundreds of swizzled addresses).
+//   * per kv tile and wave: S^T = mfma(K, Q) (2 x D/16 MFMAs),
+//     online-softmax update (lane-local m/l in exp2 domain),
+//     P->fragments in-register (pack + permlane32_swap),
+//     O^T += mfma(V^T, P^T) (2 x 2 x D/32 MFMAs).
+//   * epilogue: o = O^T / l (fp32 out), lse = ln2*(m2 + log2(l)).
+
+#include "attn_common.h"
+#include "../../include/burst_attn_hip.h"
+
+#include <stdio.h>
+
+namespace {
+
+constexpr int KVBLK = 64;
+constexpr int NTHREADS = 512;  // 8 waves
+
+template <typename T, int D>
+__global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
+    const T* __restrict__ q, const T* __restrict__ k, const T* __restrict__ v,
+    float* __restrict__ o, float* __restrict__ lse,
+    int Sq, int Sk, int N,
+    int64_t q_sb, int64_t q_ss, int64_t q_sh,
+    int64_t k_sb, int64_t k_ss, int64_t k_sh,
+    int64_t v_sb, int64_t v_ss, int64_t v_sh,
+    float scale, int causal) {
+  using MT = mfma_traits<T>;
+  using frag = typename MT::frag;
+  constexpr int SWZ_K = (D == 128) ? 15 : 7;  // K image rows are 2*D bytes
+  constexpr int SWZ_V = 7;                    // V^T image rows are 128 bytes
+  constexpr int PT = (KVBLK * D / 8) / NTHREADS;
+  static_assert(PT >= 1, "tile must fill at least one chunk per thread");
+
+  // single LDS object: [2 buffers][K row-major | V transposed][KVBLK*D]
+  __shared__ T lds[2 * 2 * KVBLK * D];
+  auto ldsK = [&](int buf) -> T* { return lds + buf * (2 * KVBLK * D); };
+  auto ldsVT = [&](int buf) -> T* {
+    return lds + buf * (2 * KVBLK * D) + KVBLK * D;
+  };
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int l31 = lane & 31;
+  const int hi = lane >> 5;
+  const int n = blockIdx.y;
+  const int b = blockIdx.z;
+  const int qb = blockIdx.x * 256 + wave * 32;
+  const int q_row = qb + l31;
+
+  const T* qp = q + (int64_t)b * q_sb + (int64_t)n * q_sh;
+  const T* kp = k + (int64_t)b * k_sb + (int64_t)n * k_sh;
+  const T* vp = v + (int64_t)b * v_sb + (int64_t)n * v_sh;
+
+  // Q fragments: B-operand of S^T = mfma(K, Q); lane holds q row q_row,
+  // elements d = 16*s + 8*hi + j  (8 contiguous -> one 16B load)
+  frag qf[D / 16];
+#pragma unroll
+  for (int s = 0; s < D / 16; ++s) {
+    if (q_row < Sq) {
+      const T* src = qp + (int64_t)q_row * q_ss + 16 * s + 8 * hi;
+      qf[s] = __builtin_bit_cast(frag, *(const u32x4_t*)src);
+    } else {
+      u32x4_t z = {0, 0, 0, 0};
+      qf[s] = __builtin_bit_cast(frag, z);
+    }
+  }
+
+  const float c2 = scale * BA_LOG2E;  // exp2-domain scale
+  float m2 = BA_NEG_BIG;
+  float lsum = 0.f;
+  f32x16_t ot[D / 32];
+#pragma unroll
+  for (int dt = 0; dt < D / 32; ++dt) ot[dt] = (f32x16_t)(0.f);
+
+  const int kv_limit = causal ? min(Sk, (int)(blockIdx.x + 1) * 256) : Sk;
+  const int nt = (kv_limit + KVBLK - 1) / KVBLK;
+
+  auto issue_loads = [&](int tile, u32x4_t* kreg, u32x4_t* vreg) {
+    const int kv0 = tile * KVBLK;
+#pragma unroll
+    for (int c = 0; c < PT; ++c) {
+      const int flat = tid + c * NTHREADS;
+      const int row = flat / (D / 8);
+      const int col8 = flat % (D / 8);
+      const int kvg = kv0 + row;
+      if (kvg < Sk) {
+        kreg[c] = *(const u32x4_t*)(kp + (int64_t)kvg * k_ss + col8 * 8);
+        vreg[c] = *(const u32x4_t*)(vp + (int64_t)kvg * v_ss + col8 * 8);
+      } else {
+        u32x4_t z = {0, 0, 0, 0};
+        kreg[c] = z;
+        vreg[c] = z;
+      }
+    }
+  };
+  auto write_lds = [&](int buf, const u32x4_t* kreg, const u32x4_t* vreg) {
+#pragma unroll
+    for (int c = 0; c < PT; ++c) {
+      const int flat = tid + c * NTHREADS;
+      const int row = flat / (D / 8);
+      const int col8 = flat % (D / 8);
+      const int byte = ba_swz<SWZ_K>(row * (2 * D) + col8 * 16, row);
+      *(u32x4_t*)((char*)ldsK(buf) + byte) = kreg[c];
+      ba_st_transposed<T, KVBLK, SWZ_V>(ldsVT(buf), row, col8 * 8, vreg[c]);
+    }
+  };
+
+  {  // prologue: tile 0
+    u32x4_t kreg[PT], vreg[PT];
+    issue_loads(0, kreg, vreg);
+    write_lds(0, kreg, vreg);
+    __syncthreads();
+  }
+
+  int cur = 0;
+  for (int t = 0; t < nt; ++t) {
+    const int kv0 = t * KVBLK;
+    const bool has_next = (t + 1) < nt;
+    u32x4_t kreg[PT], vreg[PT];
+    if (has_next) issue_loads(t + 1, kreg, vreg);
+
+    const bool active = !causal || (kv0 <= qb + 31);
+    if (active) {
+      // ---- S^T = mfma(K, Q): two 32-kv subtiles
+      f32x16_t st0 = (f32x16_t)(0.f), st1 = (f32x16_t)(0.f);
+#pragma unroll
+      for (int s = 0; s < D / 16; ++s) {
+        frag k0 = ba_ld_rowslice<T, D, SWZ_K>(ldsK(cur), l31, 16 * s + 8 * hi);
+        frag k1 =
+            ba_ld_rowslice<T, D, SWZ_K>(ldsK(cur), 32 + l31, 16 * s + 8 * hi);
+        st0 = MT::mma(k0, qf[s], st0);
+        st1 = MT::mma(k1, qf[s], st1);
+      }
+      // ---- scale into exp2 domain + mask
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int kv_g0 = kv0 + ba_crow(r, 0) + 4 * hi;
+        const int kv_g1 = kv_g0 + 32;
+        st0[r] = (kv_g0 < Sk && (!causal || kv_g0 <= q_row)) ? st0[r] * c2
+                                                             : BA_NEG_BIG;
+        st1[r] = (kv_g1 < Sk && (!causal || kv_g1 <= q_row)) ? st1[r] * c2
+                                                             : BA_NEG_BIG;
+      }
+      // ---- online softmax update (lane-local)
+      float tm = BA_NEG_BIG;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) tm = fmaxf(tm, fmaxf(st0[r], st1[r]));
+      tm = fmaxf(tm, __shfl_xor(tm, 32));
+      const float mnew = fmaxf(m2, tm);
+      const float alpha = exp2f(m2 - mnew);
+      m2 = mnew;
+      float rowsum = 0.f;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        st0[r] = exp2f(st0[r] - mnew);
+        st1[r] = exp2f(st1[r] - mnew);
+        rowsum += st0[r] + st1[r];
+      }
+      rowsum += __shfl_xor(rowsum, 32);
+      lsum = lsum * alpha + rowsum;
+#pragma unroll
+      for (int dt = 0; dt < D / 32; ++dt)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) ot[dt][r] *= alpha;
+
+      // ---- P -> fragments, O^T += mfma(V^T, P^T)
+      frag pf0[2], pf1[2];
+      ba_build_frag_pair<T>(st0, pf0);
+      ba_build_frag_pair<T>(st1, pf1);
+#pragma unroll
+      for (int dt = 0; dt < D / 32; ++dt) {
+        const int drow = dt * 32 + l31;
+#pragma unroll
+        for (int u = 0; u < 2; ++u) {
+          frag v0 = ba_ld_rowslice<T, KVBLK, SWZ_V>(ldsVT(cur), drow,
+                                                    16 * u + 8 * hi);
+          ot[dt] = MT::mma(v0, pf0[u], ot[dt]);
+          frag v1 = ba_ld_rowslice<T, KVBLK, SWZ_V>(ldsVT(cur), drow,
+                                                    32 + 16 * u + 8 * hi);
+          ot[dt] = MT::mma(v1, pf1[u], ot[dt]);
+        }
+      }
+    }
+
+    if (has_next) write_lds(cur ^ 1, kreg, vreg);
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  // ---- epilogue
+  if (q_row < Sq) {
+    const float inv_l = 1.f / lsum;
+    float* orow = o + (((int64_t)b * Sq + q_row) * N + n) * D;
+#pragma unroll
+    for (int dt = 0; dt < D / 32; ++dt)
+#pragma unroll
+      for (int r = 0; r < 16; ++r)
+        orow[dt * 32 + ba_crow(r, hi)] = ot[dt][r] * inv_l;
+    if (hi == 0)
+      lse[((int64_t)b * N + n) * Sq + q_row] = BA_LN2 * (m2 + log2f(lsum));
+  }
+}
+
+// ---- MFMA layout probe (test support): D = A*B for one 32x32x16 tile,
+// with A,B,D moved per the layout assumptions above.  a: [32][16] row-major,
+// b: [16][32] row-major, d: [32][32] row-major, all dense in HBM.
+template <typename T>
+__global__ void mfma_probe_kernel(const T* a, const T* b, float* d) {
+  using MT = mfma_traits<T>;
+  using frag = typename MT::frag;
+  const int lane = threadIdx.x & 63;
+  const int l31 = lane & 31, hi = lane >> 5;
+  frag af, bf;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    af[j] = a[l31 * 16 + 8 * hi + j];    // A[row=l31][k=8*hi+j]
+    bf[j] = b[(8 * hi + j) * 32 + l31];  // B[k=8*hi+j][col=l31]
+  }
+  f32x16_t c = (f32x16_t)(0.f);
+  c = MT::mma(af, bf, c);
+#pragma unroll
+  for (int r = 0; r < 16; ++r) d[ba_crow(r, hi) * 32 + l31] = c[r];
+}
+
+}  // namespace
+
+static thread_local char g_err[256] = "";
+extern "C" const char* bahip_last_error(void) { return g_err; }
+
+template <typename T, int D>
+static int launch_fwd(const void* q, const void* k, const void* v, float* o,
+                      float* lse, int64_t B, int64_t Sq, int64_t Sk, int64_t N,
+                      const int64_t* qs, const int64_t* ks, const int64_t* vs,
+                      float scale, int causal, void* stream) {
+  dim3 grid((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B);
+  attn_fwd_kernel<T, D><<<grid, dim3(NTHREADS), 0, (hipStream_t)stream>>>(
+      (const T*)q, (const T*)k, (const T*)v, o, lse, (int)Sq, (int)Sk, (int)N,
+      qs[0], qs[1], qs[2], ks[0], ks[1], ks[2], vs[0], vs[1], vs[2], scale,
+      causal);
+  BA_CHECK_LAUNCH();
+  return 0;
+}
+
+extern "C" int bahip_attn_fwd(const void* q, const void* k, const void* v,
+                              float* o, float* lse, int64_t B, int64_t Sq,
+                              int64_t Sk, int64_t N, int64_t D,
+                              const int64_t q_strides[3],
+                              const int64_t k_strides[3],
+                              const int64_t v_strides[3], float softmax_scale,
+                              int causal, int dtype, void* stream) {
+  if (causal && Sq != Sk) {
+    snprintf(g_err, sizeof g_err, "causal tiles require Sq == Sk (%d vs %d)",
+             (int)Sq, (int)Sk);
+    return 1001;
+  }
+  if (D == 128) {
+    if (dtype == BAHIP_BF16)
+      return launch_fwd<__bf16, 128>(q, k, v, o, lse, B, Sq, Sk, N, q_strides,
+                                     k_strides, v_strides, softmax_scale,
+                                     causal, stream);
+    if (dtype == BAHIP_F16)
+      return launch_fwd<_Float16, 128>(q, k, v, o, lse, B, Sq, Sk, N,
+                                       q_strides, k_strides, v_strides,
+                                       softmax_scale, causal, stream);
+  } else if (D == 64) {
+    if (dtype == BAHIP_BF16)
+      return launch_fwd<__bf16, 64>(q, k, v, o, lse, B, Sq, Sk, N, q_strides,
+                                    k_strides, v_strides, softmax_scale,
+                                    causal, stream);
+    if (dtype == BAHIP_F16)
+      return launch_fwd<_Float16, 64>(q, k, v, o, lse, B, Sq, Sk, N, q_strides,
+                                      k_strides, v_strides, softmax_scale,
+                                      causal, stream);
+  }
+  snprintf(g_err, sizeof g_err, "unsupported head_dim %d / dtype %d", (int)D,
+           dtype);
+  return 1002;
+}
+
+extern "C" int bahip_mfma_probe(const void* a, const void* b, float* d,
+                                int dtype, void* stream) {
+  if (dtype == BAHIP_BF16)
+    mfma_probe_kernel<__bf16><<<1, 64, 0, (hipStream_t)stream>>>(
+        (const __bf16*)a, (const __bf16*)b, d);
+  else
+    mfma_probe_kernel<_Float16><<<1, 64, 0, (hipStream_t)stream>>>(
+        (const _Float16*)a, (const _Float16*)b, d);
+  BA_CHECK_LAUNCH();
+  return 0;
+}

@@ -1,0 +1,150 @@
+// Torch bindings for the BurstAttention gfx950 kernels.
+//
+// Thin layer only: shape/stride checks, output allocation, current-stream
+// plumbing.  All compute goes through the C ABI in
+// include/burst_attn_hip.h (see that header for the reference call sites
+// each entry point replaces).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "../../include/burst_attn_hip.h"
+
+extern "C" int bahip_mfma_probe(const void*, const void*, float*, int, void*);
+
+namespace {
+
+int dtype_code(const at::Tensor& t) {
+  if (t.scalar_type() == at::kHalf) return BAHIP_F16;
+  if (t.scalar_type() == at::kBFloat16) return BAHIP_BF16;
+  TORCH_CHECK(false, "burst_attn: expected fp16 or bf16, got ", t.scalar_type());
+  return -1;
+}
+
+void check_qkv(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be a GPU tensor");
+  TORCH_CHECK(t.dim() == 4, name, " must be [B,S,N,D]");
+  TORCH_CHECK(t.stride(3) == 1, name, " head_dim must be contiguous");
+  TORCH_CHECK(t.stride(1) % 8 == 0, name,
+              " seq stride must be a multiple of 8 elements (16B alignment)");
+  TORCH_CHECK(t.stride(2) % 8 == 0, name,
+              " head stride must be a multiple of 8 elements");
+}
+
+void fill_strides(const at::Tensor& t, int64_t s[3]) {
+  s[0] = t.stride(0);
+  s[1] = t.stride(1);
+  s[2] = t.stride(2);
+}
+
+#define BA_CALL(expr)                                                      \
+  do {                                                                     \
+    int rc_ = (expr);                                                      \
+    TORCH_CHECK(rc_ == 0, "burst_attn HIP kernel failed (rc=", rc_, "): ", \
+                bahip_last_error());                                       \
+  } while (0)
+
+std::vector<at::Tensor> attn_fwd(const at::Tensor& q, const at::Tensor& k,
+                                 const at::Tensor& v, double softmax_scale,
+                                 bool causal) {
+  check_qkv(q, "q");
+  check_qkv(k, "k");
+  check_qkv(v, "v");
+  const auto B = q.size(0), Sq = q.size(1), N = q.size(2), D = q.size(3);
+  const auto Sk = k.size(1);
+  TORCH_CHECK(k.size(0) == B && v.size(0) == B, "batch mismatch");
+  TORCH_CHECK(k.size(2) == N && v.size(2) == N, "heads mismatch");
+  TORCH_CHECK(k.size(3) == D && v.size(3) == D, "head_dim mismatch");
+  TORCH_CHECK(v.size(1) == Sk, "k/v seqlen mismatch");
+  TORCH_CHECK(q.scalar_type() == k.scalar_type() &&
+                  q.scalar_type() == v.scalar_type(),
+              "q/k/v dtype mismatch");
+  auto o = at::empty({B, Sq, N, D}, q.options().dtype(at::kFloat));
+  auto lse = at::empty({B, N, Sq}, q.options().dtype(at::kFloat));
+  int64_t qs[3], ks[3], vs[3];
+  fill_strides(q, qs);
+  fill_strides(k, ks);
+  fill_strides(v, vs);
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  BA_CALL(bahip_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                         o.data_ptr<float>(), lse.data_ptr<float>(), B, Sq,
+                         Sk, N, D, qs, ks, vs, (float)softmax_scale,
+                         causal ? 1 : 0, dtype_code(q), stream));
+  return {o, lse};
+}
+
+at::Tensor attn_bwd_preprocess(const at::Tensor& o, const at::Tensor& dout) {
+  check_qkv(o, "o");
+  check_qkv(dout, "dout");
+  const auto B = o.size(0), S = o.size(1), N = o.size(2), D = o.size(3);
+  TORCH_CHECK(dout.sizes() == o.sizes(), "o/dout shape mismatch");
+  auto delta = at::empty({B, N, S}, o.options().dtype(at::kFloat));
+  int64_t os[3], gs[3];
+  fill_strides(o, os);
+  fill_strides(dout, gs);
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  BA_CALL(bahip_attn_bwd_preprocess(o.data_ptr(), dout.data_ptr(),
+                                    delta.data_ptr<float>(), B, S, N, D, os,
+                                    gs, dtype_code(o), dtype_code(dout),
+                                    stream));
+  return delta;
+}
+
+std::vector<at::Tensor> attn_bwd(const at::Tensor& dout, const at::Tensor& q,
+                                 const at::Tensor& k, const at::Tensor& v,
+                                 const at::Tensor& delta,
+                                 const at::Tensor& lse, double softmax_scale,
+                                 bool causal, bool deterministic) {
+  check_qkv(dout, "dout");
+  check_qkv(q, "q");
+  check_qkv(k, "k");
+  check_qkv(v, "v");
+  const auto B = q.size(0), Sq = q.size(1), N = q.size(2), D = q.size(3);
+  const auto Sk = k.size(1);
+  TORCH_CHECK(delta.scalar_type() == at::kFloat && lse.scalar_type() == at::kFloat,
+              "delta/lse must be fp32");
+  TORCH_CHECK(delta.dim() == 3 && lse.dim() == 3, "delta/lse must be [B,N,S]");
+  TORCH_CHECK(delta.stride(2) == 1 && lse.stride(2) == 1,
+              "delta/lse seq dim must be contiguous");
+  TORCH_CHECK(delta.size(2) == Sq && lse.size(2) == Sq,
+              "delta/lse seqlen mismatch");
+  auto dq = at::empty({B, Sq, N, D}, q.options().dtype(at::kFloat));
+  auto dk = at::empty({B, Sk, N, D}, q.options().dtype(at::kFloat));
+  auto dv = at::empty({B, Sk, N, D}, q.options().dtype(at::kFloat));
+  int64_t gs[3], qs[3], ks[3], vs[3];
+  fill_strides(dout, gs);
+  fill_strides(q, qs);
+  fill_strides(k, ks);
+  fill_strides(v, vs);
+  int64_t ds[2] = {delta.stride(0), delta.stride(1)};
+  int64_t ls[2] = {lse.stride(0), lse.stride(1)};
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  BA_CALL(bahip_attn_bwd(dout.data_ptr(), q.data_ptr(), k.data_ptr(),
+                         v.data_ptr(), delta.data_ptr<float>(),
+                         lse.data_ptr<float>(), dq.data_ptr<float>(),
+                         dk.data_ptr<float>(), dv.data_ptr<float>(), B, Sq,
+                         Sk, N, D, gs, qs, ks, vs, ds, ls,
+                         (float)softmax_scale, causal ? 1 : 0,
+                         deterministic ? 1 : 0, dtype_code(q), stream));
+  return {dq, dk, dv};
+}
+
+at::Tensor mfma_probe(const at::Tensor& a, const at::Tensor& b) {
+  TORCH_CHECK(a.is_cuda() && b.is_cuda() && a.is_contiguous() && b.is_contiguous());
+  TORCH_CHECK(a.sizes() == at::IntArrayRef({32, 16}) &&
+              b.sizes() == at::IntArrayRef({16, 32}));
+  auto d = at::zeros({32, 32}, a.options().dtype(at::kFloat));
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  BA_CALL(bahip_mfma_probe(a.data_ptr(), b.data_ptr(), d.data_ptr<float>(),
+                           dtype_code(a), stream));
+  return d;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("attn_fwd", &attn_fwd, "BurstAttention fwd tile (gfx950)");
+  m.def("attn_bwd_preprocess", &attn_bwd_preprocess, "delta = rowsum(o*do)");
+  m.def("attn_bwd", &attn_bwd, "BurstAttention bwd tile (gfx950)");
+  m.def("mfma_probe", &mfma_probe, "32x32x16 MFMA layout probe");
+}

@@ -1,0 +1,199 @@
+"""GPU parity tests: gfx950 HIP kernels vs the CPU oracle.
+
+All marked @pytest.mark.gpu (run via gpurun / the round-end driver on a real
+MI355X).  The oracle is computed live on CPU from the same seeded inputs —
+nothing here reads /root/reference.
+
+Tolerances: inputs are fp16/bf16, kernel accumulates fp32 (matching the
+flash path the reference uses); the reference's own parity gate is
+rtol=1e-3, atol=1e-2 at fp16 (test/checker.py:10).  bf16 inputs get a
+looser atol (3x fewer mantissa bits).
+"""
+
+import math
+
+import pytest
+import torch
+
+import oracle
+
+pytestmark = pytest.mark.gpu
+
+TOL = {
+    torch.float16: dict(rtol=2e-3, atol=1e-2),
+    torch.bfloat16: dict(rtol=2e-2, atol=5e-2),
+}
+BWD_TOL = {
+    torch.float16: dict(rtol=5e-3, atol=2e-2),
+    torch.bfloat16: dict(rtol=3e-2, atol=1e-1),
+}
+
+
+def _ext():
+    from burst_attn_amd._ext import load_extension
+
+    return load_extension()
+
+
+def _rand(b, s, n, d, dtype, seed):
+    g = torch.Generator().manual_seed(seed)
+    return torch.randn(b, s, n, d, generator=g).to(dtype).cuda()
+
+
+@pytest.mark.parametrize("dtype", [torch.float16, torch.bfloat16])
+def test_mfma_layout_probe(dtype):
+    """Pin the assumed 32x32x16 MFMA A/B/D lane maps with an ASYMMETRIC
+    matmul (a symmetric one passes transposed layouts silently)."""
+    g = torch.Generator().manual_seed(99)
+    a = torch.randn(32, 16, generator=g).to(dtype).cuda()
+    b = (torch.randn(16, 32, generator=g) * torch.linspace(0.2, 2.0, 32)).to(dtype).cuda()
+    d = _ext().mfma_probe(a, b)
+    ref = a.float().cpu() @ b.float().cpu()
+    torch.testing.assert_close(d.cpu(), ref, rtol=1e-2, atol=1e-2)
+
+
+@pytest.mark.parametrize("dtype", [torch.float16, torch.bfloat16])
+@pytest.mark.parametrize("causal", [False, True])
+@pytest.mark.parametrize(
+    "b,sq,sk,n,d",
+    [
+        (1, 256, 256, 2, 128),
+        (2, 512, 512, 3, 128),
+        (1, 256, 256, 2, 64),
+        (1, 96, 320, 2, 128),   # uneven, sq != sk (non-causal only)
+        (1, 300, 300, 1, 128),  # ragged tails
+    ],
+)
+def test_fwd_tile_parity(b, sq, sk, n, d, causal, dtype):
+    if causal and sq != sk:
+        pytest.skip("causal tiles are equal-length")
+    q = _rand(b, sq, n, d, dtype, 1)
+    k = _rand(b, sk, n, d, dtype, 2)
+    v = _rand(b, sk, n, d, dtype, 3)
+    scale = 1.0 / math.sqrt(d)
+    o, lse = _ext().attn_fwd(q, k, v, scale, causal)
+    o_ref, lse_ref = oracle.tile_fwd(q.cpu(), k.cpu(), v.cpu(), scale, causal)
+    torch.testing.assert_close(o.cpu(), o_ref, **TOL[dtype])
+    torch.testing.assert_close(lse.cpu(), lse_ref, rtol=1e-3, atol=2e-2)
+
+
+@pytest.mark.parametrize("dtype", [torch.float16])
+def test_fwd_strided_slices(dtype):
+    """The ring passes sliced views (zigzag halves / striped shifts) —
+    exercise non-contiguous seq slices through the stride path."""
+    b, s, n, d = 2, 256, 2, 128
+    q = _rand(b, s, n, d, dtype, 7)
+    k = _rand(b, s, n, d, dtype, 8)
+    v = _rand(b, s, n, d, dtype, 9)
+    half = s // 2
+    scale = 1.0 / math.sqrt(d)
+    # q second half vs full kv (zigzag round), via views not copies
+    o, lse = _ext().attn_fwd(q[:, half:], k, v, scale, False)
+    o_ref, lse_ref = oracle.tile_fwd(q.cpu()[:, half:], k.cpu(), v.cpu(), scale, False)
+    torch.testing.assert_close(o.cpu(), o_ref, **TOL[dtype])
+    # striped shift: q[1:] vs k[:-1] (odd offsets exercise alignment)...
+    # shift by one keeps 16B alignment because N*D is a multiple of 8
+    o2, _ = _ext().attn_fwd(q[:, 1:], k[:, :-1], v[:, :-1], scale, True)
+    o2_ref, _ = oracle.tile_fwd(q.cpu()[:, 1:], k.cpu()[:, :-1], v.cpu()[:, :-1], scale, True)
+    torch.testing.assert_close(o2.cpu(), o2_ref, **TOL[dtype])
+
+
+@pytest.mark.parametrize("dtype", [torch.float16, torch.bfloat16])
+def test_bwd_preprocess(dtype):
+    b, s, n, d = 2, 192, 3, 128
+    o = _rand(b, s, n, d, dtype, 11)
+    do = _rand(b, s, n, d, dtype, 12)
+    delta = _ext().attn_bwd_preprocess(o, do)
+    ref = (o.float() * do.float()).sum(-1).transpose(1, 2).cpu()
+    torch.testing.assert_close(delta.cpu(), ref, rtol=1e-3, atol=1e-2)
+
+
+@pytest.mark.parametrize("dtype", [torch.float16, torch.bfloat16])
+@pytest.mark.parametrize("causal", [False, True])
+@pytest.mark.parametrize(
+    "b,sq,sk,n,d",
+    [
+        (1, 256, 256, 2, 128),
+        (1, 512, 512, 2, 128),
+        (1, 256, 256, 2, 64),
+        (1, 128, 320, 2, 128),  # sq != sk
+        (1, 200, 200, 1, 128),  # ragged
+    ],
+)
+def test_bwd_tile_parity(b, sq, sk, n, d, causal, dtype):
+    if causal and sq != sk:
+        pytest.skip("causal tiles are equal-length")
+    q = _rand(b, sq, n, d, dtype, 21)
+    k = _rand(b, sk, n, d, dtype, 22)
+    v = _rand(b, sk, n, d, dtype, 23)
+    do = _rand(b, sq, n, d, dtype, 24)
+    scale = 1.0 / math.sqrt(d)
+    ext = _ext()
+    o, lse = ext.attn_fwd(q, k, v, scale, causal)
+    delta = ext.attn_bwd_preprocess(o.to(dtype), do)
+    dq, dk, dv = ext.attn_bwd(do, q, k, v, delta, lse, scale, causal, False)
+    # oracle with its own fp32 forward
+    o_ref, lse_ref = oracle.tile_fwd(q.cpu(), k.cpu(), v.cpu(), scale, causal)
+    dq_r, dk_r, dv_r = oracle.tile_bwd(
+        do.cpu(), q.cpu(), k.cpu(), v.cpu(), lse_ref, scale, causal, o=o_ref
+    )
+    torch.testing.assert_close(dv.cpu(), dv_r, **BWD_TOL[dtype])
+    torch.testing.assert_close(dk.cpu(), dk_r, **BWD_TOL[dtype])
+    torch.testing.assert_close(dq.cpu(), dq_r, **BWD_TOL[dtype])
+
+
+@pytest.mark.parametrize("dtype", [torch.float16])
+def test_merge_two_rounds_equals_full(dtype):
+    """Two half-kv tiles merged on GPU (the per-round LSE merge the ring
+    does) must equal one full tile — the invariant of
+    burst_attn_interface.py:214-242."""
+    from burst_attn_amd.tile import HipTileProvider
+
+    P = HipTileProvider()
+    b, s, n, d = 1, 256, 2, 128
+    q = _rand(b, s, n, d, dtype, 31)
+    k = _rand(b, 2 * s, n, d, dtype, 32)
+    v = _rand(b, 2 * s, n, d, dtype, 33)
+    scale = 1.0 / math.sqrt(d)
+    o1, lse1 = P.fwd(q, k[:, :s], v[:, :s], scale, False)
+    o2, lse2 = P.fwd(q, k[:, s:], v[:, s:], scale, False)
+    o = o1.to(torch.float32)
+    lse = lse1.transpose(-2, -1).unsqueeze(-1).contiguous()
+    o, lse = P.merge(o, lse, o2, lse2)
+    o_full, _ = P.fwd(q, k, v, scale, False)
+    torch.testing.assert_close(o, o_full, rtol=2e-3, atol=1e-2)
+
+
+@pytest.mark.parametrize("striped", [False, True])
+@pytest.mark.parametrize("causal", [False, True])
+def test_interface_end_to_end_single_rank(causal, striped):
+    """Full burst_attn_func autograd cycle on one GPU (W=1 ring) vs the
+    eager full-attention oracle."""
+    import torch.distributed as dist
+
+    if not dist.is_initialized():
+        import os
+
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29712")
+        dist.init_process_group("nccl", rank=0, world_size=1)
+    from burst_attn_amd import burst_attn_func, burst_attn_func_striped
+
+    b, s, n, d = 2, 512, 2, 128
+    dtype = torch.float16
+    q = _rand(b, s, n, d, dtype, 41).requires_grad_()
+    k = _rand(b, s, n, d, dtype, 42).requires_grad_()
+    v = _rand(b, s, n, d, dtype, 43).requires_grad_()
+    do = _rand(b, s, n, d, dtype, 44)
+    func = burst_attn_func_striped if striped else burst_attn_func
+    o = func(q, k, v, None, "cuda", causal)
+    dq, dk, dv = torch.autograd.grad(o, (q, k, v), do)
+    o_ref, dq_r, dk_r, dv_r = oracle.ring_forward_backward_reference(
+        q.detach().cpu(), k.detach().cpu(), v.detach().cpu(), do.cpu(), None, causal
+    )
+    tol = dict(rtol=2e-3, atol=1e-2)
+    btol = dict(rtol=5e-3, atol=2e-2)
+    torch.testing.assert_close(o.float().cpu(), o_ref, **tol)
+    torch.testing.assert_close(dv.float().cpu(), dv_r, **btol)
+    torch.testing.assert_close(dk.float().cpu(), dk_r, **btol)
+    torch.testing.assert_close(dq.float().cpu(), dq_r, **btol)
