@@ -139,3 +139,27 @@ def test_partition_roundtrip():
     for kw in ({"zigzag": False}, {"zigzag": True}, {"striped": True}):
         chunks = [get_chunk(t, 1, r, 4, **kw) for r in range(4)]
         torch.testing.assert_close(unchunk(chunks, 1, **kw), t)
+
+
+@pytest.mark.parametrize("causal,striped", [
+    (False, False), (True, False), (True, True),
+])
+def test_ring_sim_cpu_oracle_provider(causal, striped):
+    """Sanity of the test-side ring simulator itself (oracle tiles, W=4)."""
+    from .cpu_tile_provider import OracleTileProvider
+    from .ring_sim import simulate_ring
+
+    W = 4
+    b, s, n, d = 1, 64 * W, 2, 32
+    g = torch.Generator().manual_seed(66)
+    q = torch.randn(b, s, n, d, generator=g)
+    k = torch.randn(b, s, n, d, generator=g)
+    v = torch.randn(b, s, n, d, generator=g)
+    do = torch.randn(b, s, n, d, generator=g)
+    P = OracleTileProvider()
+    o, dq, dk, dv = simulate_ring(P, q, k, v, do, W, 1.0 / math.sqrt(d), causal, striped)
+    o_ref, dq_r, dk_r, dv_r = oracle.ring_forward_backward_reference(q, k, v, do, None, causal)
+    torch.testing.assert_close(o, o_ref, rtol=1e-4, atol=1e-4)
+    torch.testing.assert_close(dq, dq_r, rtol=1e-4, atol=1e-4)
+    torch.testing.assert_close(dk, dk_r, rtol=1e-4, atol=1e-4)
+    torch.testing.assert_close(dv, dv_r, rtol=1e-4, atol=1e-4)
