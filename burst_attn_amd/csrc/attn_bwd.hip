@@ -216,15 +216,21 @@ __global__ __launch_bounds__(512) void bwd_dq_kernel(
   }
 }
 
-// ====================== dk/dv kernel (kv-resident) ====================
+// ============== dk and dv kernels (kv-resident, q/do streamed) =========
+// Two single-purpose 8-wave kernels instead of one fused dk+dv kernel:
+// the fused form needs 4 x 64 accumulator registers per lane and spilled
+// (~90 VGPRs) at the 2-waves/SIMD budget; split, each kernel is register
+// -clean and runs at the dq kernel's rate.  Cost: S is recomputed (the
+// backward executes 8 tile GEMMs total vs flash-attn's 5).
 constexpr int QBLK = 64;
 
-template <typename T, int D>
-__global__ __launch_bounds__(256) void bwd_dkdv_kernel(
+// MODE: 0 = dV (dv^T += mfma(dO^T, P)); 1 = dK (dk^T += mfma(Q^T, dS))
+template <typename T, int D, int MODE>
+__global__ __launch_bounds__(512) void bwd_dkv_kernel(
     const T* __restrict__ dout, const T* __restrict__ q,
     const T* __restrict__ k, const T* __restrict__ v,
     const float* __restrict__ delta, const float* __restrict__ lse,
-    float* __restrict__ dk, float* __restrict__ dv, int Sq, int Sk, int N,
+    float* __restrict__ dout_acc, int Sq, int Sk, int N,
     int64_t g_sb, int64_t g_ss, int64_t g_sh,
     int64_t q_sb, int64_t q_ss, int64_t q_sh,
     int64_t k_sb, int64_t k_ss, int64_t k_sh,
@@ -235,57 +241,65 @@ __global__ __launch_bounds__(256) void bwd_dkdv_kernel(
   using frag = typename MT::frag;
   constexpr int SWZ = (D == 128) ? 15 : 7;
   constexpr int SWZ_T = 7;
-  constexpr int NT = 256;  // 4 waves -> 1 wave/SIMD, accs live in AGPRs
+  constexpr int NT = 512;
   constexpr int PT = (QBLK * D / 8) / NT;
+  // images per buffer: MODE_DV: [Q row-major | dO^T]; MODE_DK:
+  // [Q row-major | dO row-major | Q^T]
+  constexpr int IMGS = MODE == 0 ? 2 : 3;
 
-  // [2 buffers][Q | dO row-major | Q^T | dO^T transposed]
-  __shared__ T lds[2 * 4 * QBLK * D];
-  auto ldsQ = [&](int buf) -> T* { return lds + buf * (4 * QBLK * D); };
-  auto ldsG = [&](int buf) -> T* { return lds + buf * (4 * QBLK * D) + QBLK * D; };
-  auto ldsQT = [&](int buf) -> T* { return lds + buf * (4 * QBLK * D) + 2 * QBLK * D; };
-  auto ldsGT = [&](int buf) -> T* { return lds + buf * (4 * QBLK * D) + 3 * QBLK * D; };
+  __shared__ T lds[2 * IMGS * QBLK * D];
+  auto ldsQ = [&](int buf) -> T* { return lds + buf * (IMGS * QBLK * D); };
+  // MODE_DV: transposed dO; MODE_DK: row-major dO
+  auto ldsG = [&](int buf) -> T* {
+    return lds + buf * (IMGS * QBLK * D) + QBLK * D;
+  };
+  auto ldsQT = [&](int buf) -> T* {
+    return lds + buf * (IMGS * QBLK * D) + 2 * QBLK * D;
+  };
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
   const int l31 = lane & 31, hi = lane >> 5;
   const int n = blockIdx.y, b = blockIdx.z;
-  const int kvb = blockIdx.x * 128 + wave * 32;
+  const int kvb = blockIdx.x * 256 + wave * 32;
   const int kv_col = kvb + l31;
 
   const T* qp = q + b * q_sb + (int64_t)n * q_sh;
   const T* gp = dout + b * g_sb + (int64_t)n * g_sh;
-  const T* kp = k + b * k_sb + (int64_t)n * k_sh;
-  const T* vp = v + b * v_sb + (int64_t)n * v_sh;
   const float* dp_ = delta + b * d_sb + n * d_sh;
   const float* lp_ = lse + b * l_sb + n * l_sh;
 
-  // resident K/V fragments: B-operand (col = kv) of S = mfma(Q, K^T)
-  frag kf[D / 16], vf[D / 16];
+  // resident K (always, for S); resident V only in MODE_DK (for dP)
+  frag kf[D / 16], vf[MODE == 1 ? D / 16 : 1];
+  {
+    const T* kp = k + b * k_sb + (int64_t)n * k_sh;
+    const T* vp = v + b * v_sb + (int64_t)n * v_sh;
 #pragma unroll
-  for (int s = 0; s < D / 16; ++s) {
-    if (kv_col < Sk) {
-      kf[s] = __builtin_bit_cast(
-          frag, *(const u32x4_t*)(kp + (int64_t)kv_col * k_ss + 16 * s + 8 * hi));
-      vf[s] = __builtin_bit_cast(
-          frag, *(const u32x4_t*)(vp + (int64_t)kv_col * v_ss + 16 * s + 8 * hi));
-    } else {
-      u32x4_t z = {0, 0, 0, 0};
-      kf[s] = __builtin_bit_cast(frag, z);
-      vf[s] = __builtin_bit_cast(frag, z);
+    for (int s = 0; s < D / 16; ++s) {
+      if (kv_col < Sk) {
+        kf[s] = __builtin_bit_cast(
+            frag,
+            *(const u32x4_t*)(kp + (int64_t)kv_col * k_ss + 16 * s + 8 * hi));
+        if (MODE == 1)
+          vf[s] = __builtin_bit_cast(
+              frag,
+              *(const u32x4_t*)(vp + (int64_t)kv_col * v_ss + 16 * s + 8 * hi));
+      } else {
+        u32x4_t z = {0, 0, 0, 0};
+        kf[s] = __builtin_bit_cast(frag, z);
+        if (MODE == 1) vf[s] = __builtin_bit_cast(frag, z);
+      }
     }
   }
   const float c2 = scale * BA_LOG2E;
 
-  f32x16_t dvt[D / 32], dkt[D / 32];
+  f32x16_t acc[D / 32];
 #pragma unroll
-  for (int dt = 0; dt < D / 32; ++dt) {
-    dvt[dt] = (f32x16_t)(0.f);
-    dkt[dt] = (f32x16_t)(0.f);
-  }
+  for (int dt = 0; dt < D / 32; ++dt) acc[dt] = (f32x16_t)(0.f);
 
-  // causal: q tiles below this block's first kv row are fully masked
-  const int t0 = causal ? (blockIdx.x * 128) / QBLK : 0;
+  // causal: q tiles wholly before this workgroup's kv rows are masked
+  const int t0 = causal ? (blockIdx.x * 256) / QBLK : 0;
   const int nt = (Sq + QBLK - 1) / QBLK;
 
   auto issue_loads = [&](int tile, u32x4_t* qreg, u32x4_t* greg) {
@@ -312,9 +326,12 @@ __global__ __launch_bounds__(256) void bwd_dkdv_kernel(
       const int row = flat / (D / 8), col8 = flat % (D / 8);
       const int byte = ba_swz<SWZ>(row * (2 * D) + col8 * 16, row);
       *(u32x4_t*)((char*)ldsQ(buf) + byte) = qreg[c];
-      *(u32x4_t*)((char*)ldsG(buf) + byte) = greg[c];
-      ba_st_transposed<T, QBLK, SWZ_T>(ldsQT(buf), row, col8 * 8, qreg[c]);
-      ba_st_transposed<T, QBLK, SWZ_T>(ldsGT(buf), row, col8 * 8, greg[c]);
+      if (MODE == 0) {
+        ba_st_transposed<T, QBLK, SWZ_T>(ldsG(buf), row, col8 * 8, greg[c]);
+      } else {
+        *(u32x4_t*)((char*)ldsG(buf) + byte) = greg[c];
+        ba_st_transposed<T, QBLK, SWZ_T>(ldsQT(buf), row, col8 * 8, qreg[c]);
+      }
     }
   };
 
@@ -332,49 +349,50 @@ __global__ __launch_bounds__(256) void bwd_dkdv_kernel(
     u32x4_t qreg[PT], greg[PT];
     if (has_next) issue_loads(t + 1, qreg, greg);
 
-    // this wave's kv rows need q >= kvb to have any unmasked element
     const bool active = !causal || (q0 + QBLK - 1 >= kvb);
     if (active) {
 #pragma unroll
       for (int qs = 0; qs < 2; ++qs) {
-        f32x16_t st = (f32x16_t)(0.f), dpt = (f32x16_t)(0.f);
+        f32x16_t st = (f32x16_t)(0.f);
+        f32x16_t dpt = (f32x16_t)(0.f);
 #pragma unroll
         for (int s = 0; s < D / 16; ++s) {
           frag qfr = ba_ld_rowslice<T, D, SWZ>(ldsQ(cur), qs * 32 + l31,
                                                16 * s + 8 * hi);
-          frag gfr = ba_ld_rowslice<T, D, SWZ>(ldsG(cur), qs * 32 + l31,
-                                               16 * s + 8 * hi);
           st = MT::mma(qfr, kf[s], st);
-          dpt = MT::mma(gfr, vf[s], dpt);
+          if (MODE == 1) {
+            frag gfr = ba_ld_rowslice<T, D, SWZ>(ldsG(cur), qs * 32 + l31,
+                                                 16 * s + 8 * hi);
+            dpt = MT::mma(gfr, vf[s], dpt);
+          }
         }
-        f32x16_t ds;
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           const int q_g = q0 + qs * 32 + ba_crow(r, 0) + 4 * hi;
           const bool valid =
               q_g < Sq && kv_col < Sk && (!causal || q_g >= kv_col);
-          // per-q-row lse/delta: broadcast loads (same addr in lane group)
           const float l2 = valid ? lp_[q_g] * BA_LOG2E : 0.f;
-          const float dl = valid ? dp_[q_g] : 0.f;
           const float e = valid ? st[r] * c2 - l2 : BA_NEG_BIG;
           const float p = exp2f(e);
-          st[r] = p;                             // P
-          ds[r] = p * (dpt[r] - dl) * scale;     // dS
+          if (MODE == 0) {
+            st[r] = p;  // P for dV
+          } else {
+            const float dl = valid ? dp_[q_g] : 0.f;
+            st[r] = p * (dpt[r] - dl) * scale;  // dS for dK
+          }
         }
-        frag pf[2], dsf[2];
-        ba_build_frag_pair<T>(st, pf);
-        ba_build_frag_pair<T>(ds, dsf);
+        frag f01[2];
+        ba_build_frag_pair<T>(st, f01);
 #pragma unroll
         for (int dt = 0; dt < D / 32; ++dt) {
-          const int col = dt * 32 + l31;
+          const int drow = dt * 32 + l31;
 #pragma unroll
           for (int u = 0; u < 2; ++u) {
-            frag gtf = ba_ld_rowslice<T, QBLK, SWZ_T>(
-                ldsGT(cur), col, qs * 32 + 16 * u + 8 * hi);
-            dvt[dt] = MT::mma(gtf, pf[u], dvt[dt]);
-            frag qtf = ba_ld_rowslice<T, QBLK, SWZ_T>(
-                ldsQT(cur), col, qs * 32 + 16 * u + 8 * hi);
-            dkt[dt] = MT::mma(qtf, dsf[u], dkt[dt]);
+            // MODE_DV: A = dO^T row-slice; MODE_DK: A = Q^T row-slice
+            const T* timg = (MODE == 0) ? ldsG(cur) : ldsQT(cur);
+            frag tf = ba_ld_rowslice<T, QBLK, SWZ_T>(
+                timg, drow, qs * 32 + 16 * u + 8 * hi);
+            acc[dt] = MT::mma(tf, f01[u], acc[dt]);
           }
         }
       }
@@ -385,15 +403,12 @@ __global__ __launch_bounds__(256) void bwd_dkdv_kernel(
   }
 
   if (kv_col < Sk) {
-    float* dvrow = dv + (((int64_t)b * Sk + kv_col) * N + n) * D;
-    float* dkrow = dk + (((int64_t)b * Sk + kv_col) * N + n) * D;
+    float* row = dout_acc + (((int64_t)b * Sk + kv_col) * N + n) * D;
 #pragma unroll
     for (int dt = 0; dt < D / 32; ++dt)
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        dvrow[dt * 32 + ba_crow(r, hi)] = dvt[dt][r];
-        dkrow[dt * 32 + ba_crow(r, hi)] = dkt[dt][r];
-      }
+      for (int r = 0; r < 16; ++r)
+        row[dt * 32 + ba_crow(r, hi)] = acc[dt][r];
   }
 }
 
@@ -436,10 +451,16 @@ static int launch_bwd(const void* dout, const void* q, const void* k,
       ks[0], ks[1], ks[2], vs[0], vs[1], vs[2], ds[0], ds[1], ls[0], ls[1],
       scale, causal);
   BA_CHECK_LAUNCH();
-  dim3 grid_kv((unsigned)((Sk + 127) / 128), (unsigned)N, (unsigned)B);
-  bwd_dkdv_kernel<T, D><<<grid_kv, 256, 0, (hipStream_t)stream>>>(
+  dim3 grid_kv((unsigned)((Sk + 255) / 256), (unsigned)N, (unsigned)B);
+  bwd_dkv_kernel<T, D, 0><<<grid_kv, 512, 0, (hipStream_t)stream>>>(
+      (const T*)dout, (const T*)q, (const T*)k, (const T*)v, delta, lse, dv,
+      (int)Sq, (int)Sk, (int)N, gs[0], gs[1], gs[2], qs[0], qs[1], qs[2],
+      ks[0], ks[1], ks[2], vs[0], vs[1], vs[2], ds[0], ds[1], ls[0], ls[1],
+      scale, causal);
+  BA_CHECK_LAUNCH();
+  bwd_dkv_kernel<T, D, 1><<<grid_kv, 512, 0, (hipStream_t)stream>>>(
       (const T*)dout, (const T*)q, (const T*)k, (const T*)v, delta, lse, dk,
-      dv, (int)Sq, (int)Sk, (int)N, gs[0], gs[1], gs[2], qs[0], qs[1], qs[2],
+      (int)Sq, (int)Sk, (int)N, gs[0], gs[1], gs[2], qs[0], qs[1], qs[2],
       ks[0], ks[1], ks[2], vs[0], vs[1], vs[2], ds[0], ds[1], ls[0], ls[1],
       scale, causal);
   BA_CHECK_LAUNCH();

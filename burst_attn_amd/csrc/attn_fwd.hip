@@ -148,15 +148,25 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
         st0 = MT::mma(k0, qf[s], st0);
         st1 = MT::mma(k1, qf[s], st1);
       }
-      // ---- scale into exp2 domain + mask
+      // ---- scale into exp2 domain (+ mask only on boundary tiles)
+      const bool tile_full =
+          (kv0 + KVBLK <= Sk) && (!causal || (kv0 + KVBLK - 1 <= qb));
+      if (tile_full) {
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int kv_g0 = kv0 + ba_crow(r, 0) + 4 * hi;
-        const int kv_g1 = kv_g0 + 32;
-        st0[r] = (kv_g0 < Sk && (!causal || kv_g0 <= q_row)) ? st0[r] * c2
-                                                             : BA_NEG_BIG;
-        st1[r] = (kv_g1 < Sk && (!causal || kv_g1 <= q_row)) ? st1[r] * c2
-                                                             : BA_NEG_BIG;
+        for (int r = 0; r < 16; ++r) {
+          st0[r] *= c2;
+          st1[r] *= c2;
+        }
+      } else {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int kv_g0 = kv0 + ba_crow(r, 0) + 4 * hi;
+          const int kv_g1 = kv_g0 + 32;
+          st0[r] = (kv_g0 < Sk && (!causal || kv_g0 <= q_row)) ? st0[r] * c2
+                                                               : BA_NEG_BIG;
+          st1[r] = (kv_g1 < Sk && (!causal || kv_g1 <= q_row)) ? st1[r] * c2
+                                                               : BA_NEG_BIG;
+        }
       }
       // ---- online softmax update (lane-local)
       float tm = BA_NEG_BIG;
