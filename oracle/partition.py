@@ -21,7 +21,7 @@ def get_chunk(t, dim, rank, world_size, zigzag=False, striped=False):
     if striped:
         s = t.shape[dim]
         assert s % world_size == 0
-        idx = torch.arange(rank, s, world_size)
+        idx = torch.arange(rank, s, world_size, device=t.device)
         return t.index_select(dim, idx).contiguous()
     if zigzag:
         splits = t.chunk(world_size * 2, dim=dim)
