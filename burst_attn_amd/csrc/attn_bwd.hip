@@ -145,7 +145,7 @@ __global__ __launch_bounds__(512) void bwd_dq_kernel(
       const int byte = ba_swz<SWZ>(row * (2 * D) + col8 * 16, row);
       *(u32x4_t*)((char*)ldsK(buf) + byte) = kreg[c];
       *(u32x4_t*)((char*)ldsV(buf) + byte) = vreg[c];
-      ba_st_transposed<T, KVBLK, SWZ_T>(ldsKT(buf), row, col8 * 8, kreg[c]);
+      ba_st_transposed<T, KVBLK, SWZ_T, 7>(ldsKT(buf), row, col8 * 8, kreg[c]);
     }
   };
 
@@ -194,7 +194,7 @@ __global__ __launch_bounds__(512) void bwd_dq_kernel(
           const int col = dt * 32 + l31;
 #pragma unroll
           for (int u = 0; u < 2; ++u) {
-            frag ktf = ba_ld_rowslice<T, KVBLK, SWZ_T>(
+            frag ktf = ba_ld_rowslice<T, KVBLK, SWZ_T, 7>(
                 ldsKT(cur), col, kvs * 32 + 16 * u + 8 * hi);
             dqt[dt] = MT::mma(ktf, dsf[u], dqt[dt]);
           }
@@ -327,10 +327,10 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
       const int byte = ba_swz<SWZ>(row * (2 * D) + col8 * 16, row);
       *(u32x4_t*)((char*)ldsQ(buf) + byte) = qreg[c];
       if (MODE == 0) {
-        ba_st_transposed<T, QBLK, SWZ_T>(ldsG(buf), row, col8 * 8, greg[c]);
+        ba_st_transposed<T, QBLK, SWZ_T, 7>(ldsG(buf), row, col8 * 8, greg[c]);
       } else {
         *(u32x4_t*)((char*)ldsG(buf) + byte) = greg[c];
-        ba_st_transposed<T, QBLK, SWZ_T>(ldsQT(buf), row, col8 * 8, qreg[c]);
+        ba_st_transposed<T, QBLK, SWZ_T, 7>(ldsQT(buf), row, col8 * 8, qreg[c]);
       }
     }
   };
@@ -390,7 +390,7 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
           for (int u = 0; u < 2; ++u) {
             // MODE_DV: A = dO^T row-slice; MODE_DK: A = Q^T row-slice
             const T* timg = (MODE == 0) ? ldsG(cur) : ldsQT(cur);
-            frag tf = ba_ld_rowslice<T, QBLK, SWZ_T>(
+            frag tf = ba_ld_rowslice<T, QBLK, SWZ_T, 7>(
                 timg, drow, qs * 32 + 16 * u + 8 * hi);
             acc[dt] = MT::mma(tf, f01[u], acc[dt]);
           }

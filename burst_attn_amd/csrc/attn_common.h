@@ -63,10 +63,14 @@ __device__ __forceinline__ constexpr int ba_crow(int r, int hi) {
   return (r & 3) + 8 * (r >> 2) + 4 * hi;
 }
 
-// LDS byte-offset swizzle for [row][D] row-major tiles (16B granules)
-template <int SWZ>
+// LDS byte-offset swizzle (16B granules).  SWZ spreads a column access
+// over slots within the row; SWZ2 adds a (row>>3) term so that the
+// transposed-image staging writes (rows 8 apart across lanes) also land
+// on distinct banks: 32-way -> ~4-way write conflicts, reads stay
+// conflict-free (hand-checked for both ds_read_b128 lane groups).
+template <int SWZ, int SWZ2 = 0>
 __device__ __forceinline__ int ba_swz(int byte, int row) {
-  return byte ^ ((row & SWZ) << 4);
+  return byte ^ ((((row & SWZ) ^ ((row >> 3) & SWZ2))) << 4);
 }
 
 // Build the two 16-deep MFMA fragments (k-slices u=0,1 of a 32-wide axis)
@@ -94,10 +98,10 @@ __device__ __forceinline__ void ba_build_frag_pair(
 
 // 16B row-slice read: 8 contiguous elements of one row of a [rows][RS]
 // row-major swizzled tile (RS = row size in elements).
-template <typename T, int RS, int SWZ>
+template <typename T, int RS, int SWZ, int SWZ2 = 0>
 __device__ __forceinline__ typename mfma_traits<T>::frag ba_ld_rowslice(
     const T* lds, int row, int elem0) {
-  int byte = ba_swz<SWZ>(row * (2 * RS) + 2 * elem0, row);
+  int byte = ba_swz<SWZ, SWZ2>(row * (2 * RS) + 2 * elem0, row);
   u32x4_t wv = *(const u32x4_t*)((const char*)lds + byte);
   return __builtin_bit_cast(typename mfma_traits<T>::frag, wv);
 }
@@ -105,14 +109,14 @@ __device__ __forceinline__ typename mfma_traits<T>::frag ba_ld_rowslice(
 // transposed staging write: scatter one 8-element row chunk (elements
 // d0..d0+7 of source row `src_row`) into a [RS_T rows][.] transposed image
 // at rows d0..d0+7, column src_row.
-template <typename T, int RS_T, int SWZ_T>
+template <typename T, int RS_T, int SWZ_T, int SWZ2_T = 7>
 __device__ __forceinline__ void ba_st_transposed(T* lds, int src_row, int d0,
                                                  const u32x4_t& chunk) {
   const T* e = (const T*)&chunk;
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
     const int row = d0 + j;
-    const int byte = ba_swz<SWZ_T>(row * (2 * RS_T) + 2 * src_row, row);
+    const int byte = ba_swz<SWZ_T, SWZ2_T>(row * (2 * RS_T) + 2 * src_row, row);
     *(T*)((char*)lds + byte) = e[j];
   }
 }

@@ -25,13 +25,13 @@
 #include "../../include/burst_attn_hip.h"
 
 #include <stdio.h>
+#include <stdlib.h>
 
 namespace {
 
-constexpr int KVBLK = 64;
 constexpr int NTHREADS = 512;  // 8 waves
 
-template <typename T, int D>
+template <typename T, int D, int KVBLK>
 __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
     const T* __restrict__ q, const T* __restrict__ k, const T* __restrict__ v,
     float* __restrict__ o, float* __restrict__ lse,
@@ -118,7 +118,7 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
       const int col8 = flat % (D / 8);
       const int byte = ba_swz<SWZ_K>(row * (2 * D) + col8 * 16, row);
       *(u32x4_t*)((char*)ldsK(buf) + byte) = kreg[c];
-      ba_st_transposed<T, KVBLK, SWZ_V>(ldsVT(buf), row, col8 * 8, vreg[c]);
+      ba_st_transposed<T, KVBLK, SWZ_V, 7>(ldsVT(buf), row, col8 * 8, vreg[c]);
     }
   };
 
@@ -128,6 +128,10 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
     write_lds(0, kreg, vreg);
     __syncthreads();
   }
+  // static priority for the younger dispatch half (T5 static form):
+  // wave-uniform condition via readfirstlane, one s_setprio, no flips
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);
 
   int cur = 0;
   for (int t = 0; t < nt; ++t) {
@@ -168,27 +172,36 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
                                                                : BA_NEG_BIG;
         }
       }
-      // ---- online softmax update (lane-local)
+      // ---- online softmax update (lane-local), defer-max (T13):
+      // skip the O/l rescale while the running max grows by <= THR
+      // (exp2 domain); P is then bounded by 2^THR instead of 1, which
+      // the fp32 accumulate absorbs.  Decision taken BEFORE this tile's
+      // P is exponentiated (the textbook-safe order); wave-uniform.
+      constexpr float DEFER_THR = 8.f;
       float tm = BA_NEG_BIG;
 #pragma unroll
       for (int r = 0; r < 16; ++r) tm = fmaxf(tm, fmaxf(st0[r], st1[r]));
       tm = fmaxf(tm, __shfl_xor(tm, 32));
-      const float mnew = fmaxf(m2, tm);
-      const float alpha = exp2f(m2 - mnew);
-      m2 = mnew;
+      const bool rescale = !__all(tm - m2 <= DEFER_THR);
+      if (rescale) {
+        const float mnew = fmaxf(m2, tm);
+        const float alpha = exp2f(m2 - mnew);
+        m2 = mnew;
+        lsum *= alpha;
+#pragma unroll
+        for (int dt = 0; dt < D / 32; ++dt)
+#pragma unroll
+          for (int r = 0; r < 16; ++r) ot[dt][r] *= alpha;
+      }
       float rowsum = 0.f;
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
-        st0[r] = exp2f(st0[r] - mnew);
-        st1[r] = exp2f(st1[r] - mnew);
+        st0[r] = exp2f(st0[r] - m2);
+        st1[r] = exp2f(st1[r] - m2);
         rowsum += st0[r] + st1[r];
       }
       rowsum += __shfl_xor(rowsum, 32);
-      lsum = lsum * alpha + rowsum;
-#pragma unroll
-      for (int dt = 0; dt < D / 32; ++dt)
-#pragma unroll
-        for (int r = 0; r < 16; ++r) ot[dt][r] *= alpha;
+      lsum += rowsum;
 
       // ---- P -> fragments, O^T += mfma(V^T, P^T)
       frag pf0[2], pf1[2];
@@ -199,10 +212,10 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
         const int drow = dt * 32 + l31;
 #pragma unroll
         for (int u = 0; u < 2; ++u) {
-          frag v0 = ba_ld_rowslice<T, KVBLK, SWZ_V>(ldsVT(cur), drow,
+          frag v0 = ba_ld_rowslice<T, KVBLK, SWZ_V, 7>(ldsVT(cur), drow,
                                                     16 * u + 8 * hi);
           ot[dt] = MT::mma(v0, pf0[u], ot[dt]);
-          frag v1 = ba_ld_rowslice<T, KVBLK, SWZ_V>(ldsVT(cur), drow,
+          frag v1 = ba_ld_rowslice<T, KVBLK, SWZ_V, 7>(ldsVT(cur), drow,
                                                     32 + 16 * u + 8 * hi);
           ot[dt] = MT::mma(v1, pf1[u], ot[dt]);
         }
@@ -259,8 +272,18 @@ static int launch_fwd(const void* q, const void* k, const void* v, float* o,
                       float* lse, int64_t B, int64_t Sq, int64_t Sk, int64_t N,
                       const int64_t* qs, const int64_t* ks, const int64_t* vs,
                       float scale, int causal, void* stream) {
+  static const int kvb = [] {
+    const char* e = getenv("BA_FWD_KVBLK");
+    return e ? atoi(e) : 64;
+  }();
   dim3 grid((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B);
-  attn_fwd_kernel<T, D><<<grid, dim3(NTHREADS), 0, (hipStream_t)stream>>>(
+  if (kvb == 128)
+    attn_fwd_kernel<T, D, 128><<<grid, dim3(NTHREADS), 0, (hipStream_t)stream>>>(
+        (const T*)q, (const T*)k, (const T*)v, o, lse, (int)Sq, (int)Sk,
+        (int)N, qs[0], qs[1], qs[2], ks[0], ks[1], ks[2], vs[0], vs[1], vs[2],
+        scale, causal);
+  else
+    attn_fwd_kernel<T, D, 64><<<grid, dim3(NTHREADS), 0, (hipStream_t)stream>>>(
       (const T*)q, (const T*)k, (const T*)v, o, lse, (int)Sq, (int)Sk, (int)N,
       qs[0], qs[1], qs[2], ks[0], ks[1], ks[2], vs[0], vs[1], vs[2], scale,
       causal);
