@@ -31,7 +31,13 @@ namespace {
 
 constexpr int NTHREADS = 512;  // 8 waves
 
-template <typename T, int D, int KVBLK>
+// OUT_STATE=0: write normalised o (fp32) + lse — the stateless tile.
+// OUT_STATE=1: carry-in/carry-out accumulator state (acc = unnormalised
+// O, m = running max in the exp2 domain, l = running sum) — the fused
+// in-kernel merge replacing the reference's per-round
+// cuda_scale_out_lse_helper pass (burst_utils.py:20-33; lao.py's
+// carry-in design, lao.py:108-114).
+template <typename T, int D, int KVBLK, int OUT_STATE>
 __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
     const T* __restrict__ q, const T* __restrict__ k, const T* __restrict__ v,
     float* __restrict__ o, float* __restrict__ lse,
@@ -39,7 +45,12 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
     int64_t q_sb, int64_t q_ss, int64_t q_sh,
     int64_t k_sb, int64_t k_ss, int64_t k_sh,
     int64_t v_sb, int64_t v_ss, int64_t v_sh,
-    float scale, int causal) {
+    float scale, int causal,
+    // state (OUT_STATE=1): acc [B,Sq,N,D] strided, m/l [B,N,Sq] strided
+    float* __restrict__ st_acc, float* __restrict__ st_m,
+    float* __restrict__ st_l,
+    int64_t a_sb, int64_t a_ss, int64_t a_sh,
+    int64_t ml_sb, int64_t ml_sh, int carry_in) {
   using MT = mfma_traits<T>;
   using frag = typename MT::frag;
   constexpr int SWZ_K = (D == 128) ? 15 : 7;  // K image rows are 2*D bytes
@@ -88,6 +99,15 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
   f32x16_t ot[D / 32];
 #pragma unroll
   for (int dt = 0; dt < D / 32; ++dt) ot[dt] = (f32x16_t)(0.f);
+  if (OUT_STATE && carry_in && q_row < Sq) {
+    m2 = st_m[b * ml_sb + n * ml_sh + q_row];
+    lsum = st_l[b * ml_sb + n * ml_sh + q_row];
+    const float* arow = st_acc + b * a_sb + (int64_t)q_row * a_ss + n * a_sh;
+#pragma unroll
+    for (int dt = 0; dt < D / 32; ++dt)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) ot[dt][r] = arow[dt * 32 + ba_crow(r, hi)];
+  }
 
   const int kv_limit = causal ? min(Sk, (int)(blockIdx.x + 1) * 256) : Sk;
   const int nt = (kv_limit + KVBLK - 1) / KVBLK;
@@ -229,16 +249,56 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
 
   // ---- epilogue
   if (q_row < Sq) {
-    const float inv_l = 1.f / lsum;
-    float* orow = o + (((int64_t)b * Sq + q_row) * N + n) * D;
+    if (OUT_STATE) {
+      float* arow = st_acc + b * a_sb + (int64_t)q_row * a_ss + n * a_sh;
 #pragma unroll
-    for (int dt = 0; dt < D / 32; ++dt)
+      for (int dt = 0; dt < D / 32; ++dt)
 #pragma unroll
-      for (int r = 0; r < 16; ++r)
-        orow[dt * 32 + ba_crow(r, hi)] = ot[dt][r] * inv_l;
-    if (hi == 0)
-      lse[((int64_t)b * N + n) * Sq + q_row] = BA_LN2 * (m2 + log2f(lsum));
+        for (int r = 0; r < 16; ++r)
+          arow[dt * 32 + ba_crow(r, hi)] = ot[dt][r];
+      if (hi == 0) {
+        st_m[b * ml_sb + n * ml_sh + q_row] = m2;
+        st_l[b * ml_sb + n * ml_sh + q_row] = lsum;
+      }
+    } else {
+      const float inv_l = 1.f / lsum;
+      float* orow = o + (((int64_t)b * Sq + q_row) * N + n) * D;
+#pragma unroll
+      for (int dt = 0; dt < D / 32; ++dt)
+#pragma unroll
+        for (int r = 0; r < 16; ++r)
+          orow[dt * 32 + ba_crow(r, hi)] = ot[dt][r] * inv_l;
+      if (hi == 0)
+        lse[((int64_t)b * N + n) * Sq + q_row] = BA_LN2 * (m2 + log2f(lsum));
+    }
   }
+}
+
+// finalize: o = acc / l (cast to T), lse = ln2 * (m + log2(l)).
+// acc/m/l are the FULL chunk (contiguous); one wave row-group per row.
+template <typename T, int D>
+__global__ void attn_fwd_finalize_kernel(
+    const float* __restrict__ acc, const float* __restrict__ m,
+    const float* __restrict__ l, T* __restrict__ o, float* __restrict__ lse,
+    int S, int N) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  constexpr int LPR = D / 16;           // lanes per row (8 for D=128)
+  const int rows_per_wave = 64 / LPR;   // 8
+  const int row_in = wave * rows_per_wave + lane / LPR;
+  const int sub = lane % LPR;
+  const int s = blockIdx.x * 4 * rows_per_wave + row_in;
+  const int n = blockIdx.y, b = blockIdx.z;
+  if (s >= S) return;
+  const float lv = l[((int64_t)b * N + n) * S + s];
+  const float inv_l = 1.f / lv;
+  const float* arow = acc + (((int64_t)b * S + s) * N + n) * D + sub * 16;
+  T* orow = o + (((int64_t)b * S + s) * N + n) * D + sub * 16;
+#pragma unroll
+  for (int j = 0; j < 16; ++j) orow[j] = (T)(arow[j] * inv_l);
+  if (sub == 0)
+    lse[((int64_t)b * N + n) * S + s] =
+        BA_LN2 * (m[((int64_t)b * N + n) * S + s] + log2f(lv));
 }
 
 // ---- MFMA layout probe (test support): D = A*B for one 32x32x16 tile,
@@ -278,15 +338,82 @@ static int launch_fwd(const void* q, const void* k, const void* v, float* o,
   }();
   dim3 grid((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B);
   if (kvb == 128)
-    attn_fwd_kernel<T, D, 128><<<grid, dim3(NTHREADS), 0, (hipStream_t)stream>>>(
+    attn_fwd_kernel<T, D, 128, 0><<<grid, dim3(NTHREADS), 0, (hipStream_t)stream>>>(
         (const T*)q, (const T*)k, (const T*)v, o, lse, (int)Sq, (int)Sk,
         (int)N, qs[0], qs[1], qs[2], ks[0], ks[1], ks[2], vs[0], vs[1], vs[2],
-        scale, causal);
+        scale, causal, nullptr, nullptr, nullptr, 0, 0, 0, 0, 0, 0);
   else
-    attn_fwd_kernel<T, D, 64><<<grid, dim3(NTHREADS), 0, (hipStream_t)stream>>>(
-      (const T*)q, (const T*)k, (const T*)v, o, lse, (int)Sq, (int)Sk, (int)N,
-      qs[0], qs[1], qs[2], ks[0], ks[1], ks[2], vs[0], vs[1], vs[2], scale,
-      causal);
+    attn_fwd_kernel<T, D, 64, 0><<<grid, dim3(NTHREADS), 0, (hipStream_t)stream>>>(
+        (const T*)q, (const T*)k, (const T*)v, o, lse, (int)Sq, (int)Sk,
+        (int)N, qs[0], qs[1], qs[2], ks[0], ks[1], ks[2], vs[0], vs[1], vs[2],
+        scale, causal, nullptr, nullptr, nullptr, 0, 0, 0, 0, 0, 0);
+  BA_CHECK_LAUNCH();
+  return 0;
+}
+
+template <typename T, int D>
+static int launch_fwd_accum(const void* q, const void* k, const void* v,
+                            int64_t B, int64_t Sq, int64_t Sk, int64_t N,
+                            const int64_t* qs, const int64_t* ks,
+                            const int64_t* vs, float scale, int causal,
+                            float* acc, float* m, float* l,
+                            const int64_t* as, const int64_t* mls,
+                            int carry_in, void* stream) {
+  dim3 grid((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B);
+  attn_fwd_kernel<T, D, 64, 1><<<grid, dim3(NTHREADS), 0, (hipStream_t)stream>>>(
+      (const T*)q, (const T*)k, (const T*)v, nullptr, nullptr, (int)Sq,
+      (int)Sk, (int)N, qs[0], qs[1], qs[2], ks[0], ks[1], ks[2], vs[0], vs[1],
+      vs[2], scale, causal, acc, m, l, as[0], as[1], as[2], mls[0], mls[1],
+      carry_in);
+  BA_CHECK_LAUNCH();
+  return 0;
+}
+
+extern "C" int bahip_attn_fwd_accum(
+    const void* q, const void* k, const void* v, int64_t B, int64_t Sq,
+    int64_t Sk, int64_t N, int64_t D, const int64_t q_strides[3],
+    const int64_t k_strides[3], const int64_t v_strides[3],
+    float softmax_scale, int causal, int dtype, float* acc, float* m,
+    float* l, const int64_t acc_strides[3], const int64_t ml_strides[2],
+    int carry_in, void* stream) {
+  if (causal && Sq != Sk) return 1001;
+  if (D == 128 && dtype == BAHIP_BF16)
+    return launch_fwd_accum<__bf16, 128>(q, k, v, B, Sq, Sk, N, q_strides,
+                                         k_strides, v_strides, softmax_scale,
+                                         causal, acc, m, l, acc_strides,
+                                         ml_strides, carry_in, stream);
+  if (D == 128 && dtype == BAHIP_F16)
+    return launch_fwd_accum<_Float16, 128>(q, k, v, B, Sq, Sk, N, q_strides,
+                                           k_strides, v_strides, softmax_scale,
+                                           causal, acc, m, l, acc_strides,
+                                           ml_strides, carry_in, stream);
+  if (D == 64 && dtype == BAHIP_BF16)
+    return launch_fwd_accum<__bf16, 64>(q, k, v, B, Sq, Sk, N, q_strides,
+                                        k_strides, v_strides, softmax_scale,
+                                        causal, acc, m, l, acc_strides,
+                                        ml_strides, carry_in, stream);
+  if (D == 64 && dtype == BAHIP_F16)
+    return launch_fwd_accum<_Float16, 64>(q, k, v, B, Sq, Sk, N, q_strides,
+                                          k_strides, v_strides, softmax_scale,
+                                          causal, acc, m, l, acc_strides,
+                                          ml_strides, carry_in, stream);
+  return 1002;
+}
+
+extern "C" int bahip_attn_fwd_finalize(const float* acc, const float* m,
+                                       const float* l, void* o, float* lse,
+                                       int64_t B, int64_t S, int64_t N,
+                                       int64_t D, int dtype, void* stream) {
+  const int rows_per_block = (D == 64) ? 64 : 32;  // 4 waves x 64/(D/16)
+  dim3 grid((unsigned)((S + rows_per_block - 1) / rows_per_block), (unsigned)N,
+            (unsigned)B);
+#define LAUNCH_FIN(T, DD)                                                   attn_fwd_finalize_kernel<T, DD><<<grid, 256, 0, (hipStream_t)stream>>>(       acc, m, l, (T*)o, lse, (int)S, (int)N)
+  if (D == 128 && dtype == BAHIP_BF16) LAUNCH_FIN(__bf16, 128);
+  else if (D == 128 && dtype == BAHIP_F16) LAUNCH_FIN(_Float16, 128);
+  else if (D == 64 && dtype == BAHIP_BF16) LAUNCH_FIN(__bf16, 64);
+  else if (D == 64 && dtype == BAHIP_F16) LAUNCH_FIN(_Float16, 64);
+  else return 1002;
+#undef LAUNCH_FIN
   BA_CHECK_LAUNCH();
   return 0;
 }

@@ -73,6 +73,53 @@ std::vector<at::Tensor> attn_fwd(const at::Tensor& q, const at::Tensor& k,
   return {o, lse};
 }
 
+void attn_fwd_accum(const at::Tensor& q, const at::Tensor& k,
+                    const at::Tensor& v, double softmax_scale, bool causal,
+                    at::Tensor& acc, at::Tensor& m, at::Tensor& l,
+                    bool carry_in) {
+  check_qkv(q, "q");
+  check_qkv(k, "k");
+  check_qkv(v, "v");
+  const auto B = q.size(0), Sq = q.size(1), N = q.size(2), D = q.size(3);
+  const auto Sk = k.size(1);
+  TORCH_CHECK(acc.scalar_type() == at::kFloat && m.scalar_type() == at::kFloat
+                  && l.scalar_type() == at::kFloat, "state must be fp32");
+  TORCH_CHECK(acc.size(1) == Sq && m.size(2) == Sq && l.size(2) == Sq,
+              "state rows must match q rows");
+  TORCH_CHECK(acc.stride(3) == 1 && m.stride(2) == 1 && l.stride(2) == 1,
+              "state innermost dims must be contiguous");
+  int64_t qs[3], ks[3], vs[3], as[3];
+  fill_strides(q, qs);
+  fill_strides(k, ks);
+  fill_strides(v, vs);
+  fill_strides(acc, as);
+  int64_t mls[2] = {m.stride(0), m.stride(1)};
+  TORCH_CHECK(l.stride(0) == mls[0] && l.stride(1) == mls[1],
+              "m/l stride mismatch");
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  BA_CALL(bahip_attn_fwd_accum(
+      q.data_ptr(), k.data_ptr(), v.data_ptr(), B, Sq, Sk, N, D, qs, ks, vs,
+      (float)softmax_scale, causal ? 1 : 0, dtype_code(q),
+      acc.data_ptr<float>(), m.data_ptr<float>(), l.data_ptr<float>(), as,
+      mls, carry_in ? 1 : 0, stream));
+}
+
+std::vector<at::Tensor> attn_fwd_finalize(const at::Tensor& acc,
+                                          const at::Tensor& m,
+                                          const at::Tensor& l,
+                                          at::ScalarType out_dtype) {
+  TORCH_CHECK(acc.is_contiguous() && m.is_contiguous() && l.is_contiguous());
+  const auto B = acc.size(0), S = acc.size(1), N = acc.size(2), D = acc.size(3);
+  auto o = at::empty({B, S, N, D}, acc.options().dtype(out_dtype));
+  auto lse = at::empty({B, N, S}, acc.options().dtype(at::kFloat));
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  BA_CALL(bahip_attn_fwd_finalize(
+      acc.data_ptr<float>(), m.data_ptr<float>(), l.data_ptr<float>(),
+      o.data_ptr(), lse.data_ptr<float>(), B, S, N, D,
+      out_dtype == at::kHalf ? BAHIP_F16 : BAHIP_BF16, stream));
+  return {o, lse};
+}
+
 at::Tensor attn_bwd_preprocess(const at::Tensor& o, const at::Tensor& dout) {
   check_qkv(o, "o");
   check_qkv(dout, "dout");
@@ -144,6 +191,10 @@ at::Tensor mfma_probe(const at::Tensor& a, const at::Tensor& b) {
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_fwd", &attn_fwd, "BurstAttention fwd tile (gfx950)");
+  m.def("attn_fwd_accum", &attn_fwd_accum,
+        "carry-in accumulator fwd tile (in-kernel LSE merge)");
+  m.def("attn_fwd_finalize", &attn_fwd_finalize,
+        "o = acc/l (cast), lse from state");
   m.def("attn_bwd_preprocess", &attn_bwd_preprocess, "delta = rowsum(o*do)");
   m.def("attn_bwd", &attn_bwd, "BurstAttention bwd tile (gfx950)");
   m.def("mfma_probe", &mfma_probe, "32x32x16 MFMA layout probe");

@@ -127,10 +127,9 @@ def _setup_ctx(ctx, q, softmax_scale, flash, causal, optimize_bwd_comm,
     return double_group
 
 
-def _finalize_fwd(ctx, q, ori_k, ori_v, o, lse):
-    out = o.to(dtype=q.dtype)
-    # lse [B,S,N,1] fp32 -> [B,N,S] contiguous (reference :250-252)
-    lse = lse.squeeze(-1).transpose(1, 2).contiguous()
+def _finalize_fwd(ctx, P, q, ori_k, ori_v, state):
+    # o in q.dtype, lse [B,N,S] fp32 (reference :250-252 semantics)
+    out, lse = P.fwd_finalize(state, q.dtype)
     ctx.save_for_backward(q, ori_k, ori_v, lse, replicate(out))
     return out
 
@@ -153,8 +152,7 @@ class OpBurstAttn(torch.autograd.Function):
         ori_k, ori_v = replicate(k), replicate(v)
         comm_bufs = [torch.empty_like(k), torch.empty_like(v)]
         half = q.shape[1] // 2
-        o = None   # fp32 [B,S,N,D]
-        lse = None  # fp32 [B,S,N,1]
+        state = None  # provider-owned carry-in accumulator (in-kernel merge)
         for r in range(1, W + 1):
             offset = get_partition_id(double_group, r)
             split_kv = offset <= rank  # kv origin precedes this rank's chunks
@@ -162,29 +160,21 @@ class OpBurstAttn(torch.autograd.Function):
                 ring.double_ring_send_recv([k, v], comm_bufs, r)
                 ring.commit()
             if r == 1 or not causal:
-                o_i, lse_i = P.fwd(q, k, v, scale, causal)
-                if o is None:
-                    o = o_i.to(torch.float32)
-                    lse = lse_i.transpose(-2, -1).unsqueeze(-1).contiguous()
-                else:
-                    o, lse = P.merge(o, lse, o_i, lse_i)
+                state = P.fwd_accum(state, q, k, v, scale, causal)
             elif split_kv:
                 # kv chunks [src, 2W-1-src] with src < rank: only chunk
                 # [src] (first half) is attended, by all local q (:225-231)
-                o_i, lse_i = P.fwd(q, k[:, :half], v[:, :half], scale, False)
-                o, lse = P.merge(o, lse, o_i, lse_i)
+                state = P.fwd_accum(state, q, k[:, :half], v[:, :half], scale, False)
             else:
                 # src > rank: only q's second half (chunk [2W-1-rank])
                 # attends, to the full received kv (:232-235)
-                o_i, lse_i = P.fwd(q[:, half:], k, v, scale, False)
-                o[:, half:], lse[:, half:] = P.merge(
-                    o[:, half:], lse[:, half:], o_i, lse_i
-                )
+                state = P.fwd_accum(state, q[:, half:], k, v, scale, False,
+                                    row_offset=half)
             if r != W:
                 kv, comm_bufs = _record_stream(*comm_bufs), [k, v]
                 k, v = kv
                 ring.wait()
-        return _finalize_fwd(ctx, q, ori_k, ori_v, o, lse)
+        return _finalize_fwd(ctx, P, q, ori_k, ori_v, state)
 
     @staticmethod
     def backward(ctx, grad_output):
@@ -293,8 +283,7 @@ class OpBurstAttnStrip(torch.autograd.Function):
         W, rank = ring.world_size, ring.rank
         ori_k, ori_v = replicate(k), replicate(v)
         comm_bufs = [torch.empty_like(k), torch.empty_like(v)]
-        o = None
-        lse = None
+        state = None
         for r in range(1, W + 1):
             offset = get_partition_id(double_group, r)
             causal_shift = offset > rank  # kv origin follows this rank
@@ -302,22 +291,17 @@ class OpBurstAttnStrip(torch.autograd.Function):
                 ring.double_ring_send_recv([k, v], comm_bufs, r)
                 ring.commit()
             if not causal_shift or not causal:
-                o_i, lse_i = P.fwd(q, k, v, scale, causal)
-                if o is None:
-                    o = o_i.to(torch.float32)
-                    lse = lse_i.transpose(-2, -1).unsqueeze(-1).contiguous()
-                else:
-                    o, lse = P.merge(o, lse, o_i, lse_i)
+                state = P.fwd_accum(state, q, k, v, scale, causal)
             else:
                 # one-token shift: striped causal vs a later rank's kv
                 # (:463-475)
-                o_i, lse_i = P.fwd(q[:, 1:], k[:, :-1], v[:, :-1], scale, causal)
-                o[:, 1:], lse[:, 1:] = P.merge(o[:, 1:], lse[:, 1:], o_i, lse_i)
+                state = P.fwd_accum(state, q[:, 1:], k[:, :-1], v[:, :-1],
+                                    scale, causal, row_offset=1)
             if r != W:
                 kv, comm_bufs = _record_stream(*comm_bufs), [k, v]
                 k, v = kv
                 ring.wait()
-        return _finalize_fwd(ctx, q, ori_k, ori_v, o, lse)
+        return _finalize_fwd(ctx, P, q, ori_k, ori_v, state)
 
     @staticmethod
     def backward(ctx, grad_output):

@@ -64,6 +64,34 @@ class HipTileProvider:
     def merge(self, o, lse, o_i, lse_i):
         return _merge_scale_out_lse(o, lse, o_i, lse_i)
 
+    # ---- carry-in accumulator path (in-kernel LSE merge; lao.py design) --
+    # state = (acc fp32 [B,S,N,D] unnormalised O, m fp32 [B,N,S] exp2-domain
+    # running max, l fp32 [B,N,S] running sum) over the rank's FULL chunk;
+    # row_offset targets the zigzag-half / striped-shift row windows.
+    def fwd_accum(self, state, q, k, v, scale, causal, row_offset=0):
+        import torch
+
+        if state is None:
+            assert row_offset == 0, "state is created by a full-row round"
+            B, S, N, D = q.shape
+            acc = torch.empty(B, S, N, D, dtype=torch.float32, device=q.device)
+            m = torch.empty(B, N, S, dtype=torch.float32, device=q.device)
+            l = torch.empty(B, N, S, dtype=torch.float32, device=q.device)
+            self._ext.attn_fwd_accum(q, k, v, float(scale), bool(causal),
+                                     acc, m, l, False)
+            return (acc, m, l)
+        acc, m, l = state
+        sq = q.shape[1]
+        sl = slice(row_offset, row_offset + sq)
+        accv, mv, lv = acc[:, sl], m[:, :, sl], l[:, :, sl]
+        self._ext.attn_fwd_accum(q, k, v, float(scale), bool(causal),
+                                 accv, mv, lv, True)
+        return state
+
+    def fwd_finalize(self, state, out_dtype):
+        acc, m, l = state
+        return self._ext.attn_fwd_finalize(acc, m, l, out_dtype)
+
 
 _provider = None
 _provider_override = None

@@ -49,6 +49,24 @@ int bahip_attn_fwd(
     float softmax_scale, int causal, int dtype,
     void* stream);
 
+/* carry-in accumulator forward (in-kernel LSE merge; replaces the
+ * per-round cuda_scale_out_lse_helper pass, burst_utils.py:20-33).
+ * State: acc fp32 [B,S,N,D] (unnormalised O), m fp32 [B,N,S] (running max,
+ * exp2 domain), l fp32 [B,N,S] (running sum); acc strided {b,s,h}, m/l
+ * strided {b,h} with contiguous seq.  carry_in=0 initialises the state. */
+int bahip_attn_fwd_accum(
+    const void* q, const void* k, const void* v,
+    int64_t B, int64_t Sq, int64_t Sk, int64_t N, int64_t D,
+    const int64_t q_strides[3], const int64_t k_strides[3],
+    const int64_t v_strides[3], float softmax_scale, int causal, int dtype,
+    float* acc, float* m, float* l, const int64_t acc_strides[3],
+    const int64_t ml_strides[2], int carry_in, void* stream);
+
+/* o = acc / l cast to dtype; lse = ln(l) + m*ln2; contiguous full chunk */
+int bahip_attn_fwd_finalize(
+    const float* acc, const float* m, const float* l, void* o, float* lse,
+    int64_t B, int64_t S, int64_t N, int64_t D, int dtype, void* stream);
+
 int bahip_attn_bwd_preprocess(
     const void* o, const void* dout, float* delta,
     int64_t B, int64_t S, int64_t N, int64_t D,

@@ -25,3 +25,20 @@ class OracleTileProvider:
 
     def merge(self, o, lse, o_i, lse_i):
         return oracle.scale_out_lse(o, lse, o_i, lse_i)
+
+    # carry-in accumulator path: state = [o fp32 [B,S,N,D], lse [B,S,N,1]]
+    # maintained with the reference merge (burst_utils.py:20-33)
+    def fwd_accum(self, state, q, k, v, scale, causal, row_offset=0):
+        o_i, lse_i = oracle.tile_fwd(q, k, v, scale, causal)
+        if state is None:
+            assert row_offset == 0
+            return [o_i.to(torch.float32),
+                    lse_i.transpose(-2, -1).unsqueeze(-1).contiguous()]
+        o, lse = state
+        sl = slice(row_offset, row_offset + q.shape[1])
+        o[:, sl], lse[:, sl] = oracle.scale_out_lse(o[:, sl], lse[:, sl], o_i, lse_i)
+        return state
+
+    def fwd_finalize(self, state, out_dtype):
+        o, lse = state
+        return o.to(out_dtype), lse.squeeze(-1).transpose(1, 2).contiguous()

@@ -225,3 +225,34 @@ def test_fwd_defer_max_rescale_forced(dtype):
     o2, _ = _ext().attn_fwd(qg, kg, vg, scale, True)
     o2_ref, _ = oracle.tile_fwd(qg.cpu(), kg.cpu(), vg.cpu(), scale, True)
     torch.testing.assert_close(o2.cpu(), o2_ref, **TOL[dtype])
+
+
+@pytest.mark.parametrize("dtype", [torch.float16, torch.bfloat16])
+def test_accum_carry_in_matches_stateless(dtype):
+    """The in-kernel carry-in merge (fwd_accum rounds + finalize) must equal
+    one stateless full tile, including the row-offset window used by the
+    zigzag q1 rounds."""
+    from burst_attn_amd.tile import HipTileProvider
+
+    P = HipTileProvider()
+    b, s, n, d = 1, 512, 2, 128
+    q = _rand(b, s, n, d, dtype, 61)
+    k = _rand(b, 2 * s, n, d, dtype, 62)
+    v = _rand(b, 2 * s, n, d, dtype, 63)
+    scale = 1.0 / math.sqrt(d)
+    st = P.fwd_accum(None, q, k[:, :s], v[:, :s], scale, False)
+    st = P.fwd_accum(st, q, k[:, s:], v[:, s:], scale, False)
+    o, lse = P.fwd_finalize(st, dtype)
+    o_full, lse_full = _ext().attn_fwd(q, k, v, scale, False)
+    torch.testing.assert_close(o.float(), o_full, **TOL[dtype])
+    torch.testing.assert_close(lse, lse_full, rtol=1e-3, atol=2e-2)
+    # row-offset window: extra kv merged only into rows [half:]
+    half = s // 2
+    st2 = P.fwd_accum(None, q, k[:, :s], v[:, :s], scale, False)
+    st2 = P.fwd_accum(st2, q[:, half:], k[:, s:], v[:, s:], scale, False,
+                      row_offset=half)
+    o2, _ = P.fwd_finalize(st2, dtype)
+    o_top, _ = _ext().attn_fwd(q[:, :half], k[:, :s], v[:, :s], scale, False)
+    o_bot, _ = _ext().attn_fwd(q[:, half:], k, v, scale, False)
+    torch.testing.assert_close(o2[:, :half].float(), o_top, **TOL[dtype])
+    torch.testing.assert_close(o2[:, half:].float(), o_bot, **TOL[dtype])
