@@ -322,6 +322,30 @@ __global__ void mfma_probe_kernel(const T* a, const T* b, float* d) {
   for (int r = 0; r < 16; ++r) d[ba_crow(r, hi) * 32 + l31] = c[r];
 }
 
+
+// ---- ds_read_tr16_b64 semantics probe (test/dev support).
+// Fills LDS with element-index pattern, issues the transpose-read with a
+// per-mode address pattern, dumps each lane's 4 returned u16s.
+typedef __attribute__((ext_vector_type(4))) short s16x4_t;
+__global__ void tr16_probe_kernel(int mode, int* out) {
+  __shared__ unsigned short lds[4096];
+  for (int i = threadIdx.x; i < 4096; i += 64) lds[i] = (unsigned short)i;
+  __syncthreads();
+  const int l = threadIdx.x & 63;
+  int e;
+  switch (mode) {
+    case 0: e = 0; break;               // uniform base
+    case 1: e = (l & 15) * 4; break;    // per-16-group column stride 4
+    case 2: e = l * 4; break;           // per-lane stride 4
+    case 3: e = (l >> 4) * 64; break;   // per-quarter base
+    default: e = (l & 15) * 4 + (l >> 4) * 256; break;
+  }
+  auto p = (__attribute__((address_space(3))) s16x4_t*)&lds[e];
+  s16x4_t v = __builtin_amdgcn_ds_read_tr16_b64_v4i16(p);
+#pragma unroll
+  for (int j = 0; j < 4; ++j) out[l * 4 + j] = (int)(unsigned short)v[j];
+}
+
 }  // namespace
 
 static thread_local char g_err[256] = "";
@@ -452,6 +476,12 @@ extern "C" int bahip_attn_fwd(const void* q, const void* k, const void* v,
   snprintf(g_err, sizeof g_err, "unsupported head_dim %d / dtype %d", (int)D,
            dtype);
   return 1002;
+}
+
+extern "C" int bahip_tr16_probe(int mode, int* out, void* stream) {
+  tr16_probe_kernel<<<1, 64, 0, (hipStream_t)stream>>>(mode, out);
+  BA_CHECK_LAUNCH();
+  return 0;
 }
 
 extern "C" int bahip_mfma_probe(const void* a, const void* b, float* d,

@@ -11,6 +11,7 @@
 #include "../../include/burst_attn_hip.h"
 
 extern "C" int bahip_mfma_probe(const void*, const void*, float*, int, void*);
+extern "C" int bahip_tr16_probe(int, int*, void*);
 
 namespace {
 
@@ -187,6 +188,13 @@ at::Tensor mfma_probe(const at::Tensor& a, const at::Tensor& b) {
   return d;
 }
 
+at::Tensor tr16_probe(int64_t mode) {
+  auto out = at::zeros({256}, at::TensorOptions().dtype(at::kInt).device(at::kCUDA));
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  BA_CALL(bahip_tr16_probe((int)mode, out.data_ptr<int>(), stream));
+  return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -198,4 +206,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_bwd_preprocess", &attn_bwd_preprocess, "delta = rowsum(o*do)");
   m.def("attn_bwd", &attn_bwd, "BurstAttention bwd tile (gfx950)");
   m.def("mfma_probe", &mfma_probe, "32x32x16 MFMA layout probe");
+  m.def("tr16_probe", &tr16_probe, "ds_read_tr16_b64 semantics probe");
 }
