@@ -121,6 +121,62 @@ __device__ __forceinline__ void ba_st_transposed(T* lds, int src_row, int d0,
   }
 }
 
+typedef __attribute__((ext_vector_type(4))) short s16x4_t;
+typedef __attribute__((ext_vector_type(2))) unsigned int u32x2_t;
+
+// ---- hardware transpose-read path (ds_read_tr16_b64, gfx950) ---------
+// Probed semantics (profiles/r01/tr16_probe.txt): with l = 16h + 4q + i,
+//   result[j](l) = lds_u16[ addr16(16h + 4j + q) + i ],  j = 0..3
+// i.e. each lane reads 8B at its own address; a 4x4 transpose happens
+// across lanes {16h+4j+q : j} -> {16h+4q+i : i}.  So an MFMA A-operand
+// fragment A[row = dbase + (l&31)][k = kv0 + 8*(l>>5) + j] comes from a
+// ROW-MAJOR [kv][D] image when each lane addresses
+//   (kv0 + 8*(l>>5) + ((l>>2)&3)) * D + dbase + 16*((l>>4)&1) + 4*(l&3)
+// Image swizzle: byte ^= (kv & 7) << 3 (8B granules) makes both the
+// tr16 reads and the b64 staging writes (near-)conflict-free.
+template <typename T, int D>
+__device__ __forceinline__ int ba_tr16_byte(int lane, int kv0, int dbase) {
+  const int kvr = kv0 + 8 * (lane >> 5) + ((lane >> 2) & 3);
+  const int e16 = kvr * D + dbase + 16 * ((lane >> 4) & 1) + 4 * (lane & 3);
+  return (2 * e16) ^ ((kvr & 7) << 3);
+}
+
+template <typename T, int D>
+__device__ __forceinline__ typename mfma_traits<T>::frag ba_ld_tr16_frag(
+    const T* lds, int lane, int kv0, int dbase) {
+  auto p0 = (__attribute__((address_space(3))) s16x4_t*)((
+      __attribute__((address_space(3))) char*)(lds) +
+      ba_tr16_byte<T, D>(lane, kv0, dbase));
+  auto p1 = (__attribute__((address_space(3))) s16x4_t*)((
+      __attribute__((address_space(3))) char*)(lds) +
+      ba_tr16_byte<T, D>(lane, kv0 + 4, dbase));
+  s16x4_t lo = __builtin_amdgcn_ds_read_tr16_b64_v4i16(p0);
+  s16x4_t hi = __builtin_amdgcn_ds_read_tr16_b64_v4i16(p1);
+  union {
+    short s[8];
+    typename mfma_traits<T>::frag f;
+  } u;
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    u.s[j] = lo[j];
+    u.s[4 + j] = hi[j];
+  }
+  return u.f;
+}
+
+// staging write for the tr16 image: one 8-elem row chunk as two swizzled
+// 8B stores (row-major + (kv&7)<<3 XOR)
+template <typename T, int D>
+__device__ __forceinline__ void ba_st_tr16row(T* lds, int kv, int d0,
+                                              const u32x4_t& chunk) {
+  const int b0 = (2 * (kv * D + d0)) ^ ((kv & 7) << 3);
+  const int b1 = (2 * (kv * D + d0 + 4)) ^ ((kv & 7) << 3);
+  u32x2_t lohalf = {chunk.x, chunk.y};
+  u32x2_t hihalf = {chunk.z, chunk.w};
+  *(u32x2_t*)((char*)lds + b0) = lohalf;
+  *(u32x2_t*)((char*)lds + b1) = hihalf;
+}
+
 #define BA_CHECK_LAUNCH()                         \
   do {                                            \
     hipError_t e_ = hipGetLastError();            \

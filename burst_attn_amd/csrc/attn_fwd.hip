@@ -37,7 +37,9 @@ constexpr int NTHREADS = 512;  // 8 waves
 // in-kernel merge replacing the reference's per-round
 // cuda_scale_out_lse_helper pass (burst_utils.py:20-33; lao.py's
 // carry-in design, lao.py:108-114).
-template <typename T, int D, int KVBLK, int OUT_STATE>
+// VPATH: 0 = V transposed image + b128 row-slice reads;
+//         1 = V row-major (tr16-swizzled) + ds_read_tr16_b64 fragments
+template <typename T, int D, int KVBLK, int OUT_STATE, int VPATH>
 __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
     const T* __restrict__ q, const T* __restrict__ k, const T* __restrict__ v,
     float* __restrict__ o, float* __restrict__ lse,
@@ -138,7 +140,10 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
       const int col8 = flat % (D / 8);
       const int byte = ba_swz<SWZ_K>(row * (2 * D) + col8 * 16, row);
       *(u32x4_t*)((char*)ldsK(buf) + byte) = kreg[c];
-      ba_st_transposed<T, KVBLK, SWZ_V, 7>(ldsVT(buf), row, col8 * 8, vreg[c]);
+      if (VPATH == 0)
+        ba_st_transposed<T, KVBLK, SWZ_V, 7>(ldsVT(buf), row, col8 * 8, vreg[c]);
+      else
+        ba_st_tr16row<T, D>(ldsVT(buf), row, col8 * 8, vreg[c]);
     }
   };
 
@@ -232,11 +237,17 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
         const int drow = dt * 32 + l31;
 #pragma unroll
         for (int u = 0; u < 2; ++u) {
-          frag v0 = ba_ld_rowslice<T, KVBLK, SWZ_V, 7>(ldsVT(cur), drow,
+          frag v0, v1;
+          if (VPATH == 0) {
+            v0 = ba_ld_rowslice<T, KVBLK, SWZ_V, 7>(ldsVT(cur), drow,
                                                     16 * u + 8 * hi);
-          ot[dt] = MT::mma(v0, pf0[u], ot[dt]);
-          frag v1 = ba_ld_rowslice<T, KVBLK, SWZ_V, 7>(ldsVT(cur), drow,
+            v1 = ba_ld_rowslice<T, KVBLK, SWZ_V, 7>(ldsVT(cur), drow,
                                                     32 + 16 * u + 8 * hi);
+          } else {
+            v0 = ba_ld_tr16_frag<T, D>(ldsVT(cur), lane, 16 * u, dt * 32);
+            v1 = ba_ld_tr16_frag<T, D>(ldsVT(cur), lane, 32 + 16 * u, dt * 32);
+          }
+          ot[dt] = MT::mma(v0, pf0[u], ot[dt]);
           ot[dt] = MT::mma(v1, pf1[u], ot[dt]);
         }
       }
@@ -356,18 +367,18 @@ static int launch_fwd(const void* q, const void* k, const void* v, float* o,
                       float* lse, int64_t B, int64_t Sq, int64_t Sk, int64_t N,
                       const int64_t* qs, const int64_t* ks, const int64_t* vs,
                       float scale, int causal, void* stream) {
-  static const int kvb = [] {
-    const char* e = getenv("BA_FWD_KVBLK");
-    return e ? atoi(e) : 64;
+  static const int vpath = [] {
+    const char* e = getenv("BA_FWD_VPATH");
+    return e ? atoi(e) : 1;  // 1 = tr16 hardware transpose reads (default)
   }();
   dim3 grid((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B);
-  if (kvb == 128)
-    attn_fwd_kernel<T, D, 128, 0><<<grid, dim3(NTHREADS), 0, (hipStream_t)stream>>>(
+  if (vpath == 0)
+    attn_fwd_kernel<T, D, 64, 0, 0><<<grid, dim3(NTHREADS), 0, (hipStream_t)stream>>>(
         (const T*)q, (const T*)k, (const T*)v, o, lse, (int)Sq, (int)Sk,
         (int)N, qs[0], qs[1], qs[2], ks[0], ks[1], ks[2], vs[0], vs[1], vs[2],
         scale, causal, nullptr, nullptr, nullptr, 0, 0, 0, 0, 0, 0);
   else
-    attn_fwd_kernel<T, D, 64, 0><<<grid, dim3(NTHREADS), 0, (hipStream_t)stream>>>(
+    attn_fwd_kernel<T, D, 64, 0, 1><<<grid, dim3(NTHREADS), 0, (hipStream_t)stream>>>(
         (const T*)q, (const T*)k, (const T*)v, o, lse, (int)Sq, (int)Sk,
         (int)N, qs[0], qs[1], qs[2], ks[0], ks[1], ks[2], vs[0], vs[1], vs[2],
         scale, causal, nullptr, nullptr, nullptr, 0, 0, 0, 0, 0, 0);
@@ -383,12 +394,23 @@ static int launch_fwd_accum(const void* q, const void* k, const void* v,
                             float* acc, float* m, float* l,
                             const int64_t* as, const int64_t* mls,
                             int carry_in, void* stream) {
+  static const int vpath = [] {
+    const char* e = getenv("BA_FWD_VPATH");
+    return e ? atoi(e) : 1;
+  }();
   dim3 grid((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B);
-  attn_fwd_kernel<T, D, 64, 1><<<grid, dim3(NTHREADS), 0, (hipStream_t)stream>>>(
-      (const T*)q, (const T*)k, (const T*)v, nullptr, nullptr, (int)Sq,
-      (int)Sk, (int)N, qs[0], qs[1], qs[2], ks[0], ks[1], ks[2], vs[0], vs[1],
-      vs[2], scale, causal, acc, m, l, as[0], as[1], as[2], mls[0], mls[1],
-      carry_in);
+  if (vpath == 0)
+    attn_fwd_kernel<T, D, 64, 1, 0><<<grid, dim3(NTHREADS), 0, (hipStream_t)stream>>>(
+        (const T*)q, (const T*)k, (const T*)v, nullptr, nullptr, (int)Sq,
+        (int)Sk, (int)N, qs[0], qs[1], qs[2], ks[0], ks[1], ks[2], vs[0],
+        vs[1], vs[2], scale, causal, acc, m, l, as[0], as[1], as[2], mls[0],
+        mls[1], carry_in);
+  else
+    attn_fwd_kernel<T, D, 64, 1, 1><<<grid, dim3(NTHREADS), 0, (hipStream_t)stream>>>(
+        (const T*)q, (const T*)k, (const T*)v, nullptr, nullptr, (int)Sq,
+        (int)Sk, (int)N, qs[0], qs[1], qs[2], ks[0], ks[1], ks[2], vs[0],
+        vs[1], vs[2], scale, causal, acc, m, l, as[0], as[1], as[2], mls[0],
+        mls[1], carry_in);
   BA_CHECK_LAUNCH();
   return 0;
 }
