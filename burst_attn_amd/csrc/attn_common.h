@@ -152,16 +152,10 @@ __device__ __forceinline__ typename mfma_traits<T>::frag ba_ld_tr16_frag(
       ba_tr16_byte<T, D>(lane, kv0 + 4, dbase));
   s16x4_t lo = __builtin_amdgcn_ds_read_tr16_b64_v4i16(p0);
   s16x4_t hi = __builtin_amdgcn_ds_read_tr16_b64_v4i16(p1);
-  union {
-    short s[8];
-    typename mfma_traits<T>::frag f;
-  } u;
-#pragma unroll
-  for (int j = 0; j < 4; ++j) {
-    u.s[j] = lo[j];
-    u.s[4 + j] = hi[j];
-  }
-  return u.f;
+  u32x2_t l2 = __builtin_bit_cast(u32x2_t, lo);
+  u32x2_t h2 = __builtin_bit_cast(u32x2_t, hi);
+  u32x4_t w = {l2[0], l2[1], h2[0], h2[1]};
+  return __builtin_bit_cast(typename mfma_traits<T>::frag, w);
 }
 
 // staging write for the tr16 image: one 8-elem row chunk as two swizzled
