@@ -369,7 +369,7 @@ static int launch_fwd(const void* q, const void* k, const void* v, float* o,
                       float scale, int causal, void* stream) {
   static const int vpath = [] {
     const char* e = getenv("BA_FWD_VPATH");
-    return e ? atoi(e) : 1;  // 1 = tr16 hardware transpose reads (default)
+    return e ? atoi(e) : 0;  // 0 = V^T image (tr16 measured -3%; see profiles/r01)
   }();
   dim3 grid((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B);
   if (vpath == 0)
@@ -396,7 +396,7 @@ static int launch_fwd_accum(const void* q, const void* k, const void* v,
                             int carry_in, void* stream) {
   static const int vpath = [] {
     const char* e = getenv("BA_FWD_VPATH");
-    return e ? atoi(e) : 1;
+    return e ? atoi(e) : 0;
   }();
   dim3 grid((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B);
   if (vpath == 0)
