@@ -155,6 +155,8 @@ __global__ __launch_bounds__(512) void bwd_dq_kernel(
     write_lds(0, kreg, vreg);
     __syncthreads();
   }
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);  // T5 static form (younger half)
 
   int cur = 0;
   for (int t = 0; t < nt; ++t) {
@@ -247,7 +249,9 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
   // [Q row-major | dO row-major | Q^T]
   constexpr int IMGS = MODE == 0 ? 2 : 3;
 
-  __shared__ T lds[2 * IMGS * QBLK * D];
+  // one LDS object (a second __shared__ forces vmcnt(0) on every ds_read
+  // — cdna guide §5 trap 4a): tail carved for per-tile lse2/delta floats
+  __shared__ T lds[2 * IMGS * QBLK * D + 2 * (2 * QBLK) * (4 / (int)sizeof(T))];
   auto ldsQ = [&](int buf) -> T* { return lds + buf * (IMGS * QBLK * D); };
   // MODE_DV: transposed dO; MODE_DK: row-major dO
   auto ldsG = [&](int buf) -> T* {
@@ -255,6 +259,10 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
   };
   auto ldsQT = [&](int buf) -> T* {
     return lds + buf * (IMGS * QBLK * D) + 2 * QBLK * D;
+  };
+  // per-buffer float tail: [0..QBLK) = lse * log2e, [QBLK..2*QBLK) = delta
+  auto ldsF = [&](int buf) -> float* {
+    return (float*)(lds + 2 * IMGS * QBLK * D) + buf * (2 * QBLK);
   };
 
   const int tid = threadIdx.x;
@@ -302,7 +310,7 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
   const int t0 = causal ? (blockIdx.x * 256) / QBLK : 0;
   const int nt = (Sq + QBLK - 1) / QBLK;
 
-  auto issue_loads = [&](int tile, u32x4_t* qreg, u32x4_t* greg) {
+  auto issue_loads = [&](int tile, u32x4_t* qreg, u32x4_t* greg, float* lsed) {
     const int q0 = tile * QBLK;
 #pragma unroll
     for (int c = 0; c < PT; ++c) {
@@ -318,8 +326,19 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
         greg[c] = z;
       }
     }
+    // lse*log2e and delta rows for this tile (first 2*QBLK threads)
+    lsed[0] = 0.f;
+    if (tid < QBLK) {
+      const int qg = q0 + tid;
+      lsed[0] = (qg < Sq) ? lp_[qg] * BA_LOG2E : 0.f;
+    } else if (tid < 2 * QBLK && MODE == 1) {
+      const int qg = q0 + tid - QBLK;
+      lsed[0] = (qg < Sq) ? dp_[qg] : 0.f;
+    }
   };
-  auto write_lds = [&](int buf, const u32x4_t* qreg, const u32x4_t* greg) {
+  auto write_lds = [&](int buf, const u32x4_t* qreg, const u32x4_t* greg,
+                       float lsed) {
+    if (tid < 2 * QBLK) ldsF(buf)[tid] = lsed;
 #pragma unroll
     for (int c = 0; c < PT; ++c) {
       const int flat = tid + c * NT;
@@ -337,17 +356,21 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
 
   {
     u32x4_t qreg[PT], greg[PT];
-    issue_loads(t0, qreg, greg);
-    write_lds(0, qreg, greg);
+    float lsed;
+    issue_loads(t0, qreg, greg, &lsed);
+    write_lds(0, qreg, greg, lsed);
     __syncthreads();
   }
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);  // T5 static form (younger half)
 
   int cur = 0;
   for (int t = t0; t < nt; ++t) {
     const int q0 = t * QBLK;
     const bool has_next = (t + 1) < nt;
     u32x4_t qreg[PT], greg[PT];
-    if (has_next) issue_loads(t + 1, qreg, greg);
+    float lsed;
+    if (has_next) issue_loads(t + 1, qreg, greg, &lsed);
 
     const bool active = !causal || (q0 + QBLK - 1 >= kvb);
     if (active) {
@@ -368,16 +391,17 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
         }
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          const int q_g = q0 + qs * 32 + ba_crow(r, 0) + 4 * hi;
+          const int q_loc = qs * 32 + ba_crow(r, 0) + 4 * hi;
+          const int q_g = q0 + q_loc;
           const bool valid =
               q_g < Sq && kv_col < Sk && (!causal || q_g >= kv_col);
-          const float l2 = valid ? lp_[q_g] * BA_LOG2E : 0.f;
+          const float l2 = ldsF(cur)[q_loc];  // lse*log2e (LDS broadcast)
           const float e = valid ? st[r] * c2 - l2 : BA_NEG_BIG;
           const float p = exp2f(e);
           if (MODE == 0) {
             st[r] = p;  // P for dV
           } else {
-            const float dl = valid ? dp_[q_g] : 0.f;
+            const float dl = ldsF(cur)[QBLK + q_loc];
             st[r] = p * (dpt[r] - dl) * scale;  // dS for dK
           }
         }
@@ -397,7 +421,7 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
         }
       }
     }
-    if (has_next) write_lds(cur ^ 1, qreg, greg);
+    if (has_next) write_lds(cur ^ 1, qreg, greg, lsed);
     __syncthreads();
     cur ^= 1;
   }
