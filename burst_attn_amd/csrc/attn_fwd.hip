@@ -39,7 +39,9 @@ constexpr int NTHREADS = 512;  // 8 waves
 // carry-in design, lao.py:108-114).
 // VPATH: 0 = V transposed image + b128 row-slice reads;
 //         1 = V row-major (tr16-swizzled) + ds_read_tr16_b64 fragments
-template <typename T, int D, int KVBLK, int OUT_STATE, int VPATH>
+// SUBT: 0 = joint softmax over the 64-kv tile; 1 = per-32-subtile online
+// updates (lets subtile-0 PV MFMAs overlap subtile-1 QK/softmax)
+template <typename T, int D, int KVBLK, int OUT_STATE, int VPATH, int SUBT>
 __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
     const T* __restrict__ q, const T* __restrict__ k, const T* __restrict__ v,
     float* __restrict__ o, float* __restrict__ lse,
@@ -166,7 +168,74 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
     if (has_next) issue_loads(t + 1, kreg, vreg);
 
     const bool active = !causal || (kv0 <= qb + 31);
-    if (active) {
+    if (active && SUBT == 1) {
+      // ---- per-subtile pipeline: {QK, softmax, PV} x 2, independent
+      // chains so the scheduler overlaps PV(0) with QK(1)
+      const bool tile_full =
+          (kv0 + KVBLK <= Sk) && (!causal || (kv0 + KVBLK - 1 <= qb));
+      constexpr float DEFER_THR = 8.f;
+#pragma unroll
+      for (int kvs = 0; kvs < 2; ++kvs) {
+        f32x16_t st = (f32x16_t)(0.f);
+#pragma unroll
+        for (int s2 = 0; s2 < D / 16; ++s2) {
+          frag kf = ba_ld_rowslice<T, D, SWZ_K>(ldsK(cur), kvs * 32 + l31,
+                                                16 * s2 + 8 * hi);
+          st = MT::mma(kf, qf[s2], st);
+        }
+        if (tile_full) {
+#pragma unroll
+          for (int r = 0; r < 16; ++r) st[r] *= c2;
+        } else {
+#pragma unroll
+          for (int r = 0; r < 16; ++r) {
+            const int kv_g = kv0 + kvs * 32 + ba_crow(r, 0) + 4 * hi;
+            st[r] = (kv_g < Sk && (!causal || kv_g <= q_row)) ? st[r] * c2
+                                                              : BA_NEG_BIG;
+          }
+        }
+        float tm = BA_NEG_BIG;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) tm = fmaxf(tm, st[r]);
+        tm = fmaxf(tm, __shfl_xor(tm, 32));
+        if (!__all(tm - m2 <= DEFER_THR)) {
+          const float mnew = fmaxf(m2, tm);
+          const float alpha = exp2f(m2 - mnew);
+          m2 = mnew;
+          lsum *= alpha;
+#pragma unroll
+          for (int dt = 0; dt < D / 32; ++dt)
+#pragma unroll
+            for (int r = 0; r < 16; ++r) ot[dt][r] *= alpha;
+        }
+        float rowsum = 0.f;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          st[r] = exp2f(st[r] - m2);
+          rowsum += st[r];
+        }
+        rowsum += __shfl_xor(rowsum, 32);
+        lsum += rowsum;
+        frag pf[2];
+        ba_build_frag_pair<T>(st, pf);
+#pragma unroll
+        for (int dt = 0; dt < D / 32; ++dt) {
+          const int drow = dt * 32 + l31;
+#pragma unroll
+          for (int u = 0; u < 2; ++u) {
+            frag vv;
+            if (VPATH == 0)
+              vv = ba_ld_rowslice<T, KVBLK, SWZ_V, 7>(ldsVT(cur), drow,
+                                                      kvs * 32 + 16 * u + 8 * hi);
+            else
+              vv = ba_ld_tr16_frag<T, D>(ldsVT(cur), lane,
+                                         kvs * 32 + 16 * u, dt * 32);
+            ot[dt] = MT::mma(vv, pf[u], ot[dt]);
+          }
+        }
+      }
+    } else if (active) {
+
       // ---- S^T = mfma(K, Q): two 32-kv subtiles
       f32x16_t st0 = (f32x16_t)(0.f), st1 = (f32x16_t)(0.f);
 #pragma unroll
@@ -371,17 +440,22 @@ static int launch_fwd(const void* q, const void* k, const void* v, float* o,
     const char* e = getenv("BA_FWD_VPATH");
     return e ? atoi(e) : 0;  // 0 = V^T image (tr16 measured -3%; see profiles/r01)
   }();
+  static const int subt = [] {
+    const char* e = getenv("BA_FWD_SUBT");
+    return e ? atoi(e) : 0;
+  }();
   dim3 grid((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B);
-  if (vpath == 0)
-    attn_fwd_kernel<T, D, 64, 0, 0><<<grid, dim3(NTHREADS), 0, (hipStream_t)stream>>>(
-        (const T*)q, (const T*)k, (const T*)v, o, lse, (int)Sq, (int)Sk,
-        (int)N, qs[0], qs[1], qs[2], ks[0], ks[1], ks[2], vs[0], vs[1], vs[2],
-        scale, causal, nullptr, nullptr, nullptr, 0, 0, 0, 0, 0, 0);
-  else
-    attn_fwd_kernel<T, D, 64, 0, 1><<<grid, dim3(NTHREADS), 0, (hipStream_t)stream>>>(
-        (const T*)q, (const T*)k, (const T*)v, o, lse, (int)Sq, (int)Sk,
-        (int)N, qs[0], qs[1], qs[2], ks[0], ks[1], ks[2], vs[0], vs[1], vs[2],
-        scale, causal, nullptr, nullptr, nullptr, 0, 0, 0, 0, 0, 0);
+#define FWD_LAUNCH(VP, ST)                                                    \
+  attn_fwd_kernel<T, D, 64, 0, VP, ST>                                        \
+      <<<grid, dim3(NTHREADS), 0, (hipStream_t)stream>>>(                     \
+          (const T*)q, (const T*)k, (const T*)v, o, lse, (int)Sq, (int)Sk,    \
+          (int)N, qs[0], qs[1], qs[2], ks[0], ks[1], ks[2], vs[0], vs[1],     \
+          vs[2], scale, causal, nullptr, nullptr, nullptr, 0, 0, 0, 0, 0, 0)
+  if (vpath == 0 && subt == 0) FWD_LAUNCH(0, 0);
+  else if (vpath == 0 && subt == 1) FWD_LAUNCH(0, 1);
+  else if (vpath == 1 && subt == 0) FWD_LAUNCH(1, 0);
+  else FWD_LAUNCH(1, 1);
+#undef FWD_LAUNCH
   BA_CHECK_LAUNCH();
   return 0;
 }
@@ -399,18 +473,22 @@ static int launch_fwd_accum(const void* q, const void* k, const void* v,
     return e ? atoi(e) : 0;
   }();
   dim3 grid((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B);
-  if (vpath == 0)
-    attn_fwd_kernel<T, D, 64, 1, 0><<<grid, dim3(NTHREADS), 0, (hipStream_t)stream>>>(
-        (const T*)q, (const T*)k, (const T*)v, nullptr, nullptr, (int)Sq,
-        (int)Sk, (int)N, qs[0], qs[1], qs[2], ks[0], ks[1], ks[2], vs[0],
-        vs[1], vs[2], scale, causal, acc, m, l, as[0], as[1], as[2], mls[0],
-        mls[1], carry_in);
-  else
-    attn_fwd_kernel<T, D, 64, 1, 1><<<grid, dim3(NTHREADS), 0, (hipStream_t)stream>>>(
-        (const T*)q, (const T*)k, (const T*)v, nullptr, nullptr, (int)Sq,
-        (int)Sk, (int)N, qs[0], qs[1], qs[2], ks[0], ks[1], ks[2], vs[0],
-        vs[1], vs[2], scale, causal, acc, m, l, as[0], as[1], as[2], mls[0],
-        mls[1], carry_in);
+  static const int subt = [] {
+    const char* e = getenv("BA_FWD_SUBT");
+    return e ? atoi(e) : 0;
+  }();
+#define FWD_ALAUNCH(VP, ST)                                                   \
+  attn_fwd_kernel<T, D, 64, 1, VP, ST>                                        \
+      <<<grid, dim3(NTHREADS), 0, (hipStream_t)stream>>>(                     \
+          (const T*)q, (const T*)k, (const T*)v, nullptr, nullptr, (int)Sq,   \
+          (int)Sk, (int)N, qs[0], qs[1], qs[2], ks[0], ks[1], ks[2], vs[0],   \
+          vs[1], vs[2], scale, causal, acc, m, l, as[0], as[1], as[2],        \
+          mls[0], mls[1], carry_in)
+  if (vpath == 0 && subt == 0) FWD_ALAUNCH(0, 0);
+  else if (vpath == 0 && subt == 1) FWD_ALAUNCH(0, 1);
+  else if (vpath == 1 && subt == 0) FWD_ALAUNCH(1, 0);
+  else FWD_ALAUNCH(1, 1);
+#undef FWD_ALAUNCH
   BA_CHECK_LAUNCH();
   return 0;
 }
