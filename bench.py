@@ -46,6 +46,8 @@ def parse_args():
     p.add_argument("--dim", type=int, default=128)
     p.add_argument("--dtype", choices=["fp16", "bf16"], default="fp16")
     p.add_argument("--causal", action="store_true")
+    p.add_argument("--optimize-bwd-comm", action="store_true",
+                   help="ring the fp32 delta instead of o in backward")
     p.add_argument("--no-cpu-baseline", action="store_true")
     p.add_argument("--no-bwd", action="store_true",
                    help="skip the fwd+bwd timing leg")
@@ -165,7 +167,8 @@ def main():
         qg = q.detach().requires_grad_()
         kg = k.detach().requires_grad_()
         vg = v.detach().requires_grad_()
-        o = burst_attn_func(qg, kg, vg, None, "cuda", causal)
+        o = burst_attn_func(qg, kg, vg, None, "cuda", causal,
+                            args.optimize_bwd_comm)
         torch.autograd.grad(o, (qg, kg, vg), do)
 
     t_fwd = time_loop(fwd_step, args.steps, args.warmup)

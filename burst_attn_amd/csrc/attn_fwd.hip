@@ -442,7 +442,7 @@ static int launch_fwd(const void* q, const void* k, const void* v, float* o,
   }();
   static const int subt = [] {
     const char* e = getenv("BA_FWD_SUBT");
-    return e ? atoi(e) : 0;
+    return e ? atoi(e) : 1;  // per-subtile softmax pipeline (+3% measured)
   }();
   dim3 grid((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B);
 #define FWD_LAUNCH(VP, ST)                                                    \
@@ -475,7 +475,7 @@ static int launch_fwd_accum(const void* q, const void* k, const void* v,
   dim3 grid((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B);
   static const int subt = [] {
     const char* e = getenv("BA_FWD_SUBT");
-    return e ? atoi(e) : 0;
+    return e ? atoi(e) : 1;  // per-subtile softmax pipeline (+3% measured)
   }();
 #define FWD_ALAUNCH(VP, ST)                                                   \
   attn_fwd_kernel<T, D, 64, 1, VP, ST>                                        \
