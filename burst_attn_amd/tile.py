@@ -69,8 +69,6 @@ class HipTileProvider:
     # running max, l fp32 [B,N,S] running sum) over the rank's FULL chunk;
     # row_offset targets the zigzag-half / striped-shift row windows.
     def fwd_accum(self, state, q, k, v, scale, causal, row_offset=0):
-        import torch
-
         if state is None:
             assert row_offset == 0, "state is created by a full-row round"
             B, S, N, D = q.shape

@@ -256,3 +256,34 @@ def test_accum_carry_in_matches_stateless(dtype):
     o_bot, _ = _ext().attn_fwd(q[:, half:], k, v, scale, False)
     torch.testing.assert_close(o2[:, :half].float(), o_top, **TOL[dtype])
     torch.testing.assert_close(o2[:, half:].float(), o_bot, **TOL[dtype])
+
+
+@pytest.mark.parametrize("opt_bwd,det", [(True, False), (False, True)])
+def test_interface_flags_single_rank(opt_bwd, det):
+    """optimize_bwd_comm / deterministic flag combinations end to end on
+    one GPU (the W>1 flavours are covered by the gloo ring tests)."""
+    import torch.distributed as dist
+
+    if not dist.is_initialized():
+        import os
+
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29713")
+        dist.init_process_group("nccl", rank=0, world_size=1)
+    from burst_attn_amd import burst_attn_func
+
+    b, s, n, d = 1, 512, 2, 128
+    dtype = torch.float16
+    q = _rand(b, s, n, d, dtype, 81).requires_grad_()
+    k = _rand(b, s, n, d, dtype, 82).requires_grad_()
+    v = _rand(b, s, n, d, dtype, 83).requires_grad_()
+    do = _rand(b, s, n, d, dtype, 84)
+    o = burst_attn_func(q, k, v, None, "cuda", True, opt_bwd, det)
+    dq, dk, dv = torch.autograd.grad(o, (q, k, v), do)
+    o_ref, dq_r, dk_r, dv_r = oracle.ring_forward_backward_reference(
+        q.detach().cpu(), k.detach().cpu(), v.detach().cpu(), do.cpu(), None, True
+    )
+    torch.testing.assert_close(o.float().cpu(), o_ref, rtol=2e-3, atol=1e-2)
+    torch.testing.assert_close(dq.float().cpu(), dq_r, rtol=5e-3, atol=2e-2)
+    torch.testing.assert_close(dk.float().cpu(), dk_r, rtol=5e-3, atol=2e-2)
+    torch.testing.assert_close(dv.float().cpu(), dv_r, rtol=5e-3, atol=2e-2)
