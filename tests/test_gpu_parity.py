@@ -287,3 +287,24 @@ def test_interface_flags_single_rank(opt_bwd, det):
     torch.testing.assert_close(dq.float().cpu(), dq_r, rtol=5e-3, atol=2e-2)
     torch.testing.assert_close(dk.float().cpu(), dk_r, rtol=5e-3, atol=2e-2)
     torch.testing.assert_close(dv.float().cpu(), dv_r, rtol=5e-3, atol=2e-2)
+
+
+def test_backward_bitwise_deterministic():
+    """The two-pass backward has no atomics — repeated runs must be
+    BITWISE identical (both deterministic=True and False)."""
+    b, s, n, d = 1, 1024, 4, 128
+    dtype = torch.float16
+    q = _rand(b, s, n, d, dtype, 91)
+    k = _rand(b, s, n, d, dtype, 92)
+    v = _rand(b, s, n, d, dtype, 93)
+    do = _rand(b, s, n, d, dtype, 94)
+    ext = _ext()
+    scale = 1.0 / math.sqrt(d)
+    o, lse = ext.attn_fwd(q, k, v, scale, True)
+    delta = ext.attn_bwd_preprocess(o.to(dtype), do)
+    outs = []
+    for det in (False, True, False):
+        outs.append(ext.attn_bwd(do, q, k, v, delta, lse, scale, True, det))
+    for i in (1, 2):
+        for a, b_ in zip(outs[0], outs[i]):
+            assert torch.equal(a, b_), "backward is not bitwise deterministic"
