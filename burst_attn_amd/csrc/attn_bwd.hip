@@ -186,7 +186,7 @@ __global__ __launch_bounds__(512) void bwd_dq_kernel(
           const bool valid =
               kv_g < Sk && (!causal || kv_g <= q_row) && q_row < Sq;
           const float e = valid ? st[r] * c2 - lse2 : BA_NEG_BIG;
-          const float p = exp2f(e);
+          const float p = ba_exp2(e);
           st[r] = p * (dpt[r] - dlt) * scale;
         }
         frag dsf[2];
@@ -397,7 +397,7 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
               q_g < Sq && kv_col < Sk && (!causal || q_g >= kv_col);
           const float l2 = ldsF(cur)[q_loc];  // lse*log2e (LDS broadcast)
           const float e = valid ? st[r] * c2 - l2 : BA_NEG_BIG;
-          const float p = exp2f(e);
+          const float p = ba_exp2(e);
           if (MODE == 0) {
             st[r] = p;  // P for dV
           } else {

@@ -26,6 +26,12 @@ typedef __attribute__((ext_vector_type(16))) float f32x16_t;
 typedef __attribute__((ext_vector_type(2))) int i32x2_t;
 typedef __attribute__((ext_vector_type(4))) unsigned int u32x4_t;
 
+// raw v_exp_f32: exp2 without libm's denormal fixup (ldexp + selects per
+// call); sub-2^-126 softmax terms flush to 0, which is exactly right here
+__device__ __forceinline__ float ba_exp2(float x) {
+  return __builtin_amdgcn_exp2f(x);
+}
+
 #define BA_LOG2E 1.44269504088896340736f
 #define BA_LN2 0.69314718055994530942f
 #define BA_NEG_BIG (-1e30f)

@@ -200,7 +200,7 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
         tm = fmaxf(tm, __shfl_xor(tm, 32));
         if (!__all(tm - m2 <= DEFER_THR)) {
           const float mnew = fmaxf(m2, tm);
-          const float alpha = exp2f(m2 - mnew);
+          const float alpha = ba_exp2(m2 - mnew);
           m2 = mnew;
           lsum *= alpha;
 #pragma unroll
@@ -211,7 +211,7 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
         float rowsum = 0.f;
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          st[r] = exp2f(st[r] - m2);
+          st[r] = ba_exp2(st[r] - m2);
           rowsum += st[r];
         }
         rowsum += __shfl_xor(rowsum, 32);
@@ -279,7 +279,7 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
       const bool rescale = !__all(tm - m2 <= DEFER_THR);
       if (rescale) {
         const float mnew = fmaxf(m2, tm);
-        const float alpha = exp2f(m2 - mnew);
+        const float alpha = ba_exp2(m2 - mnew);
         m2 = mnew;
         lsum *= alpha;
 #pragma unroll
@@ -290,8 +290,8 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
       float rowsum = 0.f;
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
-        st0[r] = exp2f(st0[r] - m2);
-        st1[r] = exp2f(st1[r] - m2);
+        st0[r] = ba_exp2(st0[r] - m2);
+        st1[r] = ba_exp2(st1[r] - m2);
         rowsum += st0[r] + st1[r];
       }
       rowsum += __shfl_xor(rowsum, 32);
