@@ -226,13 +226,15 @@ __global__ __launch_bounds__(512) void bwd_dq_kernel(
 // backward executes 8 tile GEMMs total vs flash-attn's 5).
 constexpr int QBLK = 64;
 
-// MODE: 0 = dV (dv^T += mfma(dO^T, P)); 1 = dK (dk^T += mfma(Q^T, dS))
+// MODE: 0 = dV (dv^T += mfma(dO^T, P)); 1 = dK (dk^T += mfma(Q^T, dS));
+//       2 = fused dK+dV (both accumulators; 8-wave; may spill — A/B)
 template <typename T, int D, int MODE>
 __global__ __launch_bounds__(512) void bwd_dkv_kernel(
     const T* __restrict__ dout, const T* __restrict__ q,
     const T* __restrict__ k, const T* __restrict__ v,
     const float* __restrict__ delta, const float* __restrict__ lse,
-    float* __restrict__ dout_acc, int Sq, int Sk, int N,
+    float* __restrict__ dout_acc, float* __restrict__ dout_acc2,
+    int Sq, int Sk, int N,
     int64_t g_sb, int64_t g_ss, int64_t g_sh,
     int64_t q_sb, int64_t q_ss, int64_t q_sh,
     int64_t k_sb, int64_t k_ss, int64_t k_sh,
@@ -246,8 +248,8 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
   constexpr int NT = 512;
   constexpr int PT = (QBLK * D / 8) / NT;
   // images per buffer: MODE_DV: [Q row-major | dO^T]; MODE_DK:
-  // [Q row-major | dO row-major | Q^T]
-  constexpr int IMGS = MODE == 0 ? 2 : 3;
+  // [Q row-major | dO row-major | Q^T]; MODE 2: all four
+  constexpr int IMGS = MODE == 0 ? 2 : (MODE == 1 ? 3 : 4);
 
   // one LDS object (a second __shared__ forces vmcnt(0) on every ds_read
   // — cdna guide §5 trap 4a): tail carved for per-tile lse2/delta floats
@@ -259,6 +261,9 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
   };
   auto ldsQT = [&](int buf) -> T* {
     return lds + buf * (IMGS * QBLK * D) + 2 * QBLK * D;
+  };
+  auto ldsGT2 = [&](int buf) -> T* {  // MODE 2 only: dO^T as 4th image
+    return lds + buf * (IMGS * QBLK * D) + 3 * QBLK * D;
   };
   // per-buffer float tail: [0..QBLK) = lse * log2e, [QBLK..2*QBLK) = delta
   auto ldsF = [&](int buf) -> float* {
@@ -278,8 +283,8 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
   const float* dp_ = delta + b * d_sb + n * d_sh;
   const float* lp_ = lse + b * l_sb + n * l_sh;
 
-  // resident K (always, for S); resident V only in MODE_DK (for dP)
-  frag kf[D / 16], vf[MODE == 1 ? D / 16 : 1];
+  // resident K (always, for S); resident V when dP is needed
+  frag kf[D / 16], vf[MODE >= 1 ? D / 16 : 1];
   {
     const T* kp = k + b * k_sb + (int64_t)n * k_sh;
     const T* vp = v + b * v_sb + (int64_t)n * v_sh;
@@ -289,22 +294,26 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
         kf[s] = __builtin_bit_cast(
             frag,
             *(const u32x4_t*)(kp + (int64_t)kv_col * k_ss + 16 * s + 8 * hi));
-        if (MODE == 1)
+        if (MODE >= 1)
           vf[s] = __builtin_bit_cast(
               frag,
               *(const u32x4_t*)(vp + (int64_t)kv_col * v_ss + 16 * s + 8 * hi));
       } else {
         u32x4_t z = {0, 0, 0, 0};
         kf[s] = __builtin_bit_cast(frag, z);
-        if (MODE == 1) vf[s] = __builtin_bit_cast(frag, z);
+        if (MODE >= 1) vf[s] = __builtin_bit_cast(frag, z);
       }
     }
   }
   const float c2 = scale * BA_LOG2E;
 
   f32x16_t acc[D / 32];
+  f32x16_t acc2[MODE == 2 ? D / 32 : 1];  // MODE 2: dV accumulator
 #pragma unroll
   for (int dt = 0; dt < D / 32; ++dt) acc[dt] = (f32x16_t)(0.f);
+  if (MODE == 2)
+#pragma unroll
+    for (int dt = 0; dt < D / 32; ++dt) acc2[dt] = (f32x16_t)(0.f);
 
   // causal: q tiles wholly before this workgroup's kv rows are masked
   const int t0 = causal ? (blockIdx.x * 256) / QBLK : 0;
@@ -331,7 +340,7 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
     if (tid < QBLK) {
       const int qg = q0 + tid;
       lsed[0] = (qg < Sq) ? lp_[qg] * BA_LOG2E : 0.f;
-    } else if (tid < 2 * QBLK && MODE == 1) {
+    } else if (tid < 2 * QBLK && MODE >= 1) {
       const int qg = q0 + tid - QBLK;
       lsed[0] = (qg < Sq) ? dp_[qg] : 0.f;
     }
@@ -350,6 +359,8 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
       } else {
         *(u32x4_t*)((char*)ldsG(buf) + byte) = greg[c];
         ba_st_transposed<T, QBLK, SWZ_T, 7>(ldsQT(buf), row, col8 * 8, qreg[c]);
+        if (MODE == 2)
+          ba_st_transposed<T, QBLK, SWZ_T, 7>(ldsGT2(buf), row, col8 * 8, greg[c]);
       }
     }
   };
@@ -383,7 +394,7 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
           frag qfr = ba_ld_rowslice<T, D, SWZ>(ldsQ(cur), qs * 32 + l31,
                                                16 * s + 8 * hi);
           st = MT::mma(qfr, kf[s], st);
-          if (MODE == 1) {
+          if (MODE >= 1) {
             frag gfr = ba_ld_rowslice<T, D, SWZ>(ldsG(cur), qs * 32 + l31,
                                                  16 * s + 8 * hi);
             dpt = MT::mma(gfr, vf[s], dpt);
@@ -400,13 +411,19 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
           const float p = ba_exp2(e);
           if (MODE == 0) {
             st[r] = p;  // P for dV
-          } else {
+          } else if (MODE == 1) {
             const float dl = ldsF(cur)[QBLK + q_loc];
             st[r] = p * (dpt[r] - dl) * scale;  // dS for dK
+          } else {
+            const float dl = ldsF(cur)[QBLK + q_loc];
+            dpt[r] = p * (dpt[r] - dl) * scale;  // dS
+            st[r] = p;                            // P
           }
         }
         frag f01[2];
         ba_build_frag_pair<T>(st, f01);
+        frag f01b[MODE == 2 ? 2 : 1];
+        if (MODE == 2) ba_build_frag_pair<T>(dpt, (frag*)f01b);
 #pragma unroll
         for (int dt = 0; dt < D / 32; ++dt) {
           const int drow = dt * 32 + l31;
@@ -416,7 +433,15 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
             const T* timg = (MODE == 0) ? ldsG(cur) : ldsQT(cur);
             frag tf = ba_ld_rowslice<T, QBLK, SWZ_T, 7>(
                 timg, drow, qs * 32 + 16 * u + 8 * hi);
-            acc[dt] = MT::mma(tf, f01[u], acc[dt]);
+            if (MODE == 2) {
+              // acc = dK (dS frags), acc2 = dV (P frags via dO^T)
+              acc[dt] = MT::mma(tf, f01b[u], acc[dt]);
+              frag gtf = ba_ld_rowslice<T, QBLK, SWZ_T, 7>(
+                  ldsGT2(cur), drow, qs * 32 + 16 * u + 8 * hi);
+              acc2[dt] = MT::mma(gtf, f01[u], acc2[dt]);
+            } else {
+              acc[dt] = MT::mma(tf, f01[u], acc[dt]);
+            }
           }
         }
       }
@@ -433,6 +458,14 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
 #pragma unroll
       for (int r = 0; r < 16; ++r)
         row[dt * 32 + ba_crow(r, hi)] = acc[dt][r];
+    if (MODE == 2) {
+      float* row2 = dout_acc2 + (((int64_t)b * Sk + kv_col) * N + n) * D;
+#pragma unroll
+      for (int dt = 0; dt < D / 32; ++dt)
+#pragma unroll
+        for (int r = 0; r < 16; ++r)
+          row2[dt * 32 + ba_crow(r, hi)] = acc2[dt][r];
+    }
   }
 }
 
@@ -478,13 +511,13 @@ static int launch_bwd(const void* dout, const void* q, const void* k,
   dim3 grid_kv((unsigned)((Sk + 255) / 256), (unsigned)N, (unsigned)B);
   bwd_dkv_kernel<T, D, 0><<<grid_kv, 512, 0, (hipStream_t)stream>>>(
       (const T*)dout, (const T*)q, (const T*)k, (const T*)v, delta, lse, dv,
-      (int)Sq, (int)Sk, (int)N, gs[0], gs[1], gs[2], qs[0], qs[1], qs[2],
+      nullptr, (int)Sq, (int)Sk, (int)N, gs[0], gs[1], gs[2], qs[0], qs[1], qs[2],
       ks[0], ks[1], ks[2], vs[0], vs[1], vs[2], ds[0], ds[1], ls[0], ls[1],
       scale, causal);
   BA_CHECK_LAUNCH();
   bwd_dkv_kernel<T, D, 1><<<grid_kv, 512, 0, (hipStream_t)stream>>>(
       (const T*)dout, (const T*)q, (const T*)k, (const T*)v, delta, lse, dk,
-      (int)Sq, (int)Sk, (int)N, gs[0], gs[1], gs[2], qs[0], qs[1], qs[2],
+      nullptr, (int)Sq, (int)Sk, (int)N, gs[0], gs[1], gs[2], qs[0], qs[1], qs[2],
       ks[0], ks[1], ks[2], vs[0], vs[1], vs[2], ds[0], ds[1], ls[0], ls[1],
       scale, causal);
   BA_CHECK_LAUNCH();
