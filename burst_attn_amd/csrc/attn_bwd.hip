@@ -185,7 +185,7 @@ __global__ __launch_bounds__(512) void bwd_dq_kernel(
           const int kv_g = kv0 + kvs * 32 + ba_crow(r, 0) + 4 * hi;
           const bool valid =
               kv_g < Sk && (!causal || kv_g <= q_row) && q_row < Sq;
-          const float e = valid ? st[r] * c2 - lse2 : BA_NEG_BIG;
+          const float e = valid ? __builtin_fmaf(st[r], c2, -lse2) : BA_NEG_BIG;
           const float p = ba_exp2(e);
           st[r] = p * (dpt[r] - dlt) * scale;
         }
@@ -407,7 +407,7 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
           const bool valid =
               q_g < Sq && kv_col < Sk && (!causal || q_g >= kv_col);
           const float l2 = ldsF(cur)[q_loc];  // lse*log2e (LDS broadcast)
-          const float e = valid ? st[r] * c2 - l2 : BA_NEG_BIG;
+          const float e = valid ? __builtin_fmaf(st[r], c2, -l2) : BA_NEG_BIG;
           const float p = ba_exp2(e);
           if (MODE == 0) {
             st[r] = p;  // P for dV

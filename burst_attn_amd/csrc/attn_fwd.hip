@@ -183,21 +183,21 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
                                                 16 * s2 + 8 * hi);
           st = MT::mma(kf, qf[s2], st);
         }
-        if (tile_full) {
-#pragma unroll
-          for (int r = 0; r < 16; ++r) st[r] *= c2;
-        } else {
+        // fold the softmax scale into the exp argument (exp2+fma): the
+        // row max is taken on RAW scores (max commutes with c2 > 0), the
+        // scale costs one multiply on the max instead of 16 per subtile
+        if (!tile_full) {
 #pragma unroll
           for (int r = 0; r < 16; ++r) {
             const int kv_g = kv0 + kvs * 32 + ba_crow(r, 0) + 4 * hi;
-            st[r] = (kv_g < Sk && (!causal || kv_g <= q_row)) ? st[r] * c2
-                                                              : BA_NEG_BIG;
+            if (!(kv_g < Sk && (!causal || kv_g <= q_row))) st[r] = BA_NEG_BIG;
           }
         }
         float tm = BA_NEG_BIG;
 #pragma unroll
         for (int r = 0; r < 16; ++r) tm = fmaxf(tm, st[r]);
         tm = fmaxf(tm, __shfl_xor(tm, 32));
+        tm *= c2;  // into the exp2 domain
         if (!__all(tm - m2 <= DEFER_THR)) {
           const float mnew = fmaxf(m2, tm);
           const float alpha = ba_exp2(m2 - mnew);
@@ -211,7 +211,7 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
         float rowsum = 0.f;
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          st[r] = ba_exp2(st[r] - m2);
+          st[r] = ba_exp2(__builtin_fmaf(st[r], c2, -m2));
           rowsum += st[r];
         }
         rowsum += __shfl_xor(rowsum, 32);
