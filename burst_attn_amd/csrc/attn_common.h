@@ -49,6 +49,10 @@ struct mfma_traits<_Float16> {
     _Float16 h = (_Float16)x;
     return (unsigned int)__builtin_bit_cast(unsigned short, h);
   }
+  // one v_cvt_pkrtz_f16_f32 instead of two converts + shift/or
+  static __device__ __forceinline__ unsigned int pack2(float lo, float hi) {
+    return __builtin_bit_cast(unsigned int, __builtin_amdgcn_cvt_pkrtz(lo, hi));
+  }
 };
 
 template <>
@@ -60,6 +64,9 @@ struct mfma_traits<__bf16> {
   static __device__ __forceinline__ unsigned int bits(float x) {
     __bf16 h = (__bf16)x;
     return (unsigned int)__builtin_bit_cast(unsigned short, h);
+  }
+  static __device__ __forceinline__ unsigned int pack2(float lo, float hi) {
+    return bits(lo) | (bits(hi) << 16);  // hipcc fuses to v_cvt_pk_bf16_f32
   }
 };
 
@@ -90,16 +97,27 @@ __device__ __forceinline__ void ba_build_frag_pair(
   using MT = mfma_traits<T>;
 #pragma unroll
   for (int u = 0; u < 2; ++u) {
-    unsigned int a0 = MT::bits(p[8 * u + 0]) | (MT::bits(p[8 * u + 1]) << 16);
-    unsigned int a1 = MT::bits(p[8 * u + 2]) | (MT::bits(p[8 * u + 3]) << 16);
-    unsigned int b0 = MT::bits(p[8 * u + 4]) | (MT::bits(p[8 * u + 5]) << 16);
-    unsigned int b1 = MT::bits(p[8 * u + 6]) | (MT::bits(p[8 * u + 7]) << 16);
+    unsigned int a0 = MT::pack2(p[8 * u + 0], p[8 * u + 1]);
+    unsigned int a1 = MT::pack2(p[8 * u + 2], p[8 * u + 3]);
+    unsigned int b0 = MT::pack2(p[8 * u + 4], p[8 * u + 5]);
+    unsigned int b1 = MT::pack2(p[8 * u + 6], p[8 * u + 7]);
     i32x2_t r0 = __builtin_amdgcn_permlane32_swap((int)a0, (int)b0, false, false);
     i32x2_t r1 = __builtin_amdgcn_permlane32_swap((int)a1, (int)b1, false, false);
     u32x4_t w = {(unsigned int)r0[0], (unsigned int)r1[0],
                  (unsigned int)r0[1], (unsigned int)r1[1]};
     out[u] = __builtin_bit_cast(typename mfma_traits<T>::frag, w);
   }
+}
+
+// 16-value max via 3-input nesting (clang fuses fmaxf chains to v_max3)
+__device__ __forceinline__ float ba_max16(const f32x16_t& v) {
+  float a = fmaxf(fmaxf(v[0], v[1]), v[2]);
+  float b = fmaxf(fmaxf(v[3], v[4]), v[5]);
+  float c = fmaxf(fmaxf(v[6], v[7]), v[8]);
+  float d = fmaxf(fmaxf(v[9], v[10]), v[11]);
+  float e = fmaxf(fmaxf(v[12], v[13]), v[14]);
+  float f = fmaxf(fmaxf(a, b), v[15]);
+  return fmaxf(fmaxf(fmaxf(c, d), e), f);
 }
 
 // 16B row-slice read: 8 contiguous elements of one row of a [rows][RS]

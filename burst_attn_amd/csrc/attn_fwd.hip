@@ -193,9 +193,7 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
             if (!(kv_g < Sk && (!causal || kv_g <= q_row))) st[r] = BA_NEG_BIG;
           }
         }
-        float tm = BA_NEG_BIG;
-#pragma unroll
-        for (int r = 0; r < 16; ++r) tm = fmaxf(tm, st[r]);
+        float tm = ba_max16(st);
         tm = fmaxf(tm, __shfl_xor(tm, 32));
         tm *= c2;  // into the exp2 domain
         if (!__all(tm - m2 <= DEFER_THR)) {
