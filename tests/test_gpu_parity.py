@@ -335,6 +335,7 @@ def test_full_size_properties():
         lse0, torch.full_like(lse0, math.log(s)), rtol=1e-4, atol=1e-3
     )
     # causal flavour: row i attends i+1 keys -> lse = ln(i+1)
-    _, lse_c = ext.attn_fwd(zq[:, :2048], k[:, :2048] * 0, scale, True)
+    _, lse_c = ext.attn_fwd(zq[:, :2048], k[:, :2048] * 0, ones[:, :2048],
+                            scale, True)
     expect = torch.log(torch.arange(1, 2049, dtype=torch.float32)).cuda()
     torch.testing.assert_close(lse_c[0, 0], expect, rtol=1e-4, atol=1e-3)
