@@ -145,7 +145,7 @@ __global__ __launch_bounds__(512) void bwd_dq_kernel(
       const int byte = ba_swz<SWZ>(row * (2 * D) + col8 * 16, row);
       *(u32x4_t*)((char*)ldsK(buf) + byte) = kreg[c];
       *(u32x4_t*)((char*)ldsV(buf) + byte) = vreg[c];
-      ba_st_transposed<T, KVBLK, SWZ_T, 7>(ldsKT(buf), row, col8 * 8, kreg[c]);
+      ba_st_tr16row<T, D>(ldsKT(buf), row, col8 * 8, kreg[c]);
     }
   };
 
@@ -196,8 +196,8 @@ __global__ __launch_bounds__(512) void bwd_dq_kernel(
           const int col = dt * 32 + l31;
 #pragma unroll
           for (int u = 0; u < 2; ++u) {
-            frag ktf = ba_ld_rowslice<T, KVBLK, SWZ_T, 7>(
-                ldsKT(cur), col, kvs * 32 + 16 * u + 8 * hi);
+            frag ktf = ba_ld_tr16_frag<T, D>(ldsKT(cur), lane,
+                                             kvs * 32 + 16 * u, dt * 32);
             dqt[dt] = MT::mma(ktf, dsf[u], dqt[dt]);
           }
         }
@@ -335,7 +335,8 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
         greg[c] = z;
       }
     }
-    // lse*log2e and delta rows for this tile (first 2*QBLK threads)
+    // lse*log2e and delta rows, interleaved {lse2, delta} per q row so
+    // the compute loop reads one 8B word per row
     lsed[0] = 0.f;
     if (tid < QBLK) {
       const int qg = q0 + tid;
@@ -347,7 +348,8 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
   };
   auto write_lds = [&](int buf, const u32x4_t* qreg, const u32x4_t* greg,
                        float lsed) {
-    if (tid < 2 * QBLK) ldsF(buf)[tid] = lsed;
+    if (tid < QBLK) ldsF(buf)[2 * tid] = lsed;
+    else if (tid < 2 * QBLK) ldsF(buf)[2 * (tid - QBLK) + 1] = lsed;
 #pragma unroll
     for (int c = 0; c < PT; ++c) {
       const int flat = tid + c * NT;
@@ -355,12 +357,12 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
       const int byte = ba_swz<SWZ>(row * (2 * D) + col8 * 16, row);
       *(u32x4_t*)((char*)ldsQ(buf) + byte) = qreg[c];
       if (MODE == 0) {
-        ba_st_transposed<T, QBLK, SWZ_T, 7>(ldsG(buf), row, col8 * 8, greg[c]);
+        ba_st_tr16row<T, D>(ldsG(buf), row, col8 * 8, greg[c]);
       } else {
         *(u32x4_t*)((char*)ldsG(buf) + byte) = greg[c];
-        ba_st_transposed<T, QBLK, SWZ_T, 7>(ldsQT(buf), row, col8 * 8, qreg[c]);
+        ba_st_tr16row<T, D>(ldsQT(buf), row, col8 * 8, qreg[c]);
         if (MODE == 2)
-          ba_st_transposed<T, QBLK, SWZ_T, 7>(ldsGT2(buf), row, col8 * 8, greg[c]);
+          ba_st_tr16row<T, D>(ldsGT2(buf), row, col8 * 8, greg[c]);
       }
     }
   };
@@ -406,18 +408,18 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
           const int q_g = q0 + q_loc;
           const bool valid =
               q_g < Sq && kv_col < Sk && (!causal || q_g >= kv_col);
-          const float l2 = ldsF(cur)[q_loc];  // lse*log2e (LDS broadcast)
-          const float e = valid ? __builtin_fmaf(st[r], c2, -l2) : BA_NEG_BIG;
+          const float2 ld2 =
+              *(const float2*)&ldsF(cur)[2 * q_loc];  // {lse2, delta}
+          const float e =
+              valid ? __builtin_fmaf(st[r], c2, -ld2.x) : BA_NEG_BIG;
           const float p = ba_exp2(e);
           if (MODE == 0) {
             st[r] = p;  // P for dV
           } else if (MODE == 1) {
-            const float dl = ldsF(cur)[QBLK + q_loc];
-            st[r] = p * (dpt[r] - dl) * scale;  // dS for dK
+            st[r] = p * (dpt[r] - ld2.y) * scale;  // dS for dK
           } else {
-            const float dl = ldsF(cur)[QBLK + q_loc];
-            dpt[r] = p * (dpt[r] - dl) * scale;  // dS
-            st[r] = p;                            // P
+            dpt[r] = p * (dpt[r] - ld2.y) * scale;  // dS
+            st[r] = p;                              // P
           }
         }
         frag f01[2];
@@ -431,13 +433,13 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
           for (int u = 0; u < 2; ++u) {
             // MODE_DV: A = dO^T row-slice; MODE_DK: A = Q^T row-slice
             const T* timg = (MODE == 0) ? ldsG(cur) : ldsQT(cur);
-            frag tf = ba_ld_rowslice<T, QBLK, SWZ_T, 7>(
-                timg, drow, qs * 32 + 16 * u + 8 * hi);
+            frag tf = ba_ld_tr16_frag<T, D>(timg, lane, qs * 32 + 16 * u,
+                                            dt * 32);
             if (MODE == 2) {
               // acc = dK (dS frags), acc2 = dV (P frags via dO^T)
               acc[dt] = MT::mma(tf, f01b[u], acc[dt]);
-              frag gtf = ba_ld_rowslice<T, QBLK, SWZ_T, 7>(
-                  ldsGT2(cur), drow, qs * 32 + 16 * u + 8 * hi);
+              frag gtf = ba_ld_tr16_frag<T, D>(ldsGT2(cur), lane,
+                                               qs * 32 + 16 * u, dt * 32);
               acc2[dt] = MT::mma(gtf, f01[u], acc2[dt]);
             } else {
               acc[dt] = MT::mma(tf, f01[u], acc[dt]);
