@@ -145,7 +145,7 @@ __global__ __launch_bounds__(512) void bwd_dq_kernel(
       const int byte = ba_swz<SWZ>(row * (2 * D) + col8 * 16, row);
       *(u32x4_t*)((char*)ldsK(buf) + byte) = kreg[c];
       *(u32x4_t*)((char*)ldsV(buf) + byte) = vreg[c];
-      ba_st_tr16row<T, D>(ldsKT(buf), row, col8 * 8, kreg[c]);
+      ba_st_transposed<T, KVBLK, SWZ_T, 7>(ldsKT(buf), row, col8 * 8, kreg[c]);
     }
   };
 
@@ -196,8 +196,8 @@ __global__ __launch_bounds__(512) void bwd_dq_kernel(
           const int col = dt * 32 + l31;
 #pragma unroll
           for (int u = 0; u < 2; ++u) {
-            frag ktf = ba_ld_tr16_frag<T, D>(ldsKT(cur), lane,
-                                             kvs * 32 + 16 * u, dt * 32);
+            frag ktf = ba_ld_rowslice<T, KVBLK, SWZ_T, 7>(
+                ldsKT(cur), col, kvs * 32 + 16 * u + 8 * hi);
             dqt[dt] = MT::mma(ktf, dsf[u], dqt[dt]);
           }
         }
