@@ -357,7 +357,7 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
       const int byte = ba_swz<SWZ>(row * (2 * D) + col8 * 16, row);
       *(u32x4_t*)((char*)ldsQ(buf) + byte) = qreg[c];
       if (MODE == 0) {
-        ba_st_tr16row<T, D>(ldsG(buf), row, col8 * 8, greg[c]);
+        ba_st_transposed<T, QBLK, SWZ_T, 7>(ldsG(buf), row, col8 * 8, greg[c]);
       } else {
         // MODE_DK keeps the scatter-write transposed image: the tr16
         // address set spills ~47 VGPRs here (measured -14% fwd+bwd)
@@ -434,13 +434,9 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
 #pragma unroll
           for (int u = 0; u < 2; ++u) {
             // MODE_DV: A = dO^T row-slice; MODE_DK: A = Q^T row-slice
-            frag tf;
-            if (MODE == 0)
-              tf = ba_ld_tr16_frag<T, D>(ldsG(cur), lane, qs * 32 + 16 * u,
-                                         dt * 32);
-            else
-              tf = ba_ld_rowslice<T, QBLK, SWZ_T, 7>(
-                  ldsQT(cur), drow, qs * 32 + 16 * u + 8 * hi);
+            const T* timg = (MODE == 0) ? ldsG(cur) : ldsQT(cur);
+            frag tf = ba_ld_rowslice<T, QBLK, SWZ_T, 7>(
+                timg, drow, qs * 32 + 16 * u + 8 * hi);
             if (MODE == 2) {
               // acc = dK (dS frags), acc2 = dV (P frags via dO^T)
               acc[dt] = MT::mma(tf, f01b[u], acc[dt]);
