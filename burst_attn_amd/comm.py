@@ -9,18 +9,19 @@ Semantics kept from the reference:
   * per-round batched isend/irecv pairs to rank+1 / from rank-1 on the ring
     (``comm.py:148-172``), with the even/odd op ordering that avoids
     P2P deadlock (``comm.py:166-171``);
-  * ``commit()`` → ``dist.batch_isend_irecv`` (``comm.py:269``);
+  * ``commit()`` -> ``dist.batch_isend_irecv`` (``comm.py:269``);
     ``wait()`` blocks the compute stream on the transfer (``comm.py:301-321``).
     On ROCm, NCCL/RCCL P2P runs on the process group's internal HIP
     streams, so the transfer overlaps the attention kernel that is queued
     on the compute stream between commit() and wait() — the same overlap
     the reference gets from its side-stream commit (``comm.py:267-283``).
-
-The two-level "double ring" (intra-node + inter-node groups,
-``comm.py:187-254``) is multi-node-only — the reference itself disables it
-single-node (``benchmarks/benchmark.py:41-44``) — and is not implemented in
-this round (out of scope until multi-node exists).
-Passing real double-ring groups raises NotImplementedError.
+  * the two-level "double ring" (intra-node + inter-node groups) follows
+    the reference's staging scheme (``comm.py:221-254``): every round hops
+    the intra-node ring; once per ``intra_size`` rounds an inter-node
+    transfer — pre-staged a full phase earlier on the inter group, so it
+    overlaps ``intra_size`` rounds of compute — is swapped in.  The
+    travelling-dq variant merges the wandering dq into the inter-node
+    buffer at each phase boundary (``comm.py:187-218``).
 """
 
 import os
@@ -91,33 +92,35 @@ def print_rank(*args, **kwargs):
 
 
 class Ring:
-    """Batched neighbour-ring P2P over one process group.
+    """Batched neighbour-ring P2P over one process group (optionally a
+    two-level intra/inter hierarchy).
 
-    Builds per-round op lists (``send_recv``), commits them in one
-    ``batch_isend_irecv`` (one RCCL group call → one xGMI link each way),
-    and ``wait()``s before the received buffers are consumed.
-
-    API mirrors the reference ``Ring`` (``comm.py:104-321``) minus the
-    bmtrain backend and (for now) the double ring.
+    Builds per-round op lists, commits them in one ``batch_isend_irecv``
+    (one RCCL group call -> one xGMI link each way), and ``wait()``s
+    before the received buffers are consumed.  API mirrors the reference
+    ``Ring`` (``comm.py:104-321``) minus the bmtrain backend.
     """
 
     def __init__(self, process_group=None, double_group=(None, None)):
-        if double_group is not None and (
-            double_group[0] is not None or double_group[1] is not None
-        ):
-            raise NotImplementedError(
-                "double-ring (intra+inter node groups) is multi-node-only and "
-                "not implemented yet; pass double_group=[None, None]"
-            )
+        dg = double_group if double_group is not None else (None, None)
         self.comm = process_group
         self.world_size = dist.get_world_size(process_group)
         self.rank = dist.get_rank(process_group)
-        self.local_group = None
-        self.local_group2 = None
-        self.intra_size = 1
-        self.inter_size = 1
+        self.local_group = dg[0]
+        self.local_group2 = dg[1]
+        self.double_ring = dg[0] is not None and dg[1] is not None
+        if self.double_ring:
+            self.intra_size = dist.get_world_size(self.local_group)
+            self.inter_size = dist.get_world_size(self.local_group2)
+        else:
+            self.intra_size = self.world_size
+            self.inter_size = 1
+        self.buffer_list = []
         self.ops = []
         self.reqs = []
+        self._inter_ops = []
+        self._inter_reqs = []
+        self._wait_inter = False
 
     def _make_ring_ops(self, src_tensors, dst_tensors, group=None):
         comm = self.comm if group is None else group
@@ -140,32 +143,105 @@ class Ring:
                 ops += [recv_op, send_op]
         return ops
 
-    def send_recv(self, tensor_list, dest_list):
+    def send_recv(self, tensor_list, dest_list, group=None):
         """Queue a ring hop: send each tensor to rank+1, receive the
         matching buffer from rank-1 (reference ``_ring_send_recv_base``,
         ``comm.py:256-257``).  A 1-rank ring is a local copy (gloo cannot
         send-to-self; semantically identical)."""
-        if self.world_size == 1:
+        comm = self.comm if group is None else group
+        if dist.get_world_size(comm) == 1:
             for src, dst in zip(tensor_list, dest_list):
                 dst.copy_(src)
             return
-        self.ops += self._make_ring_ops(tensor_list, dest_list)
+        self.ops += self._make_ring_ops(tensor_list, dest_list, group)
 
-    # name kept so call sites read like the reference's
-    # (double_ring_send_recv degenerates to the plain hop single-node,
-    #  comm.py:221-227)
+    @staticmethod
+    def _buffers_match(bufs, tensors):
+        return len(bufs) == len(tensors) and all(
+            b.size() == t.size() and b.dtype == t.dtype
+            for b, t in zip(bufs, tensors)
+        )
+
     def double_ring_send_recv(self, tensor_list, dest_list, r=0):
-        self.send_recv(tensor_list, dest_list)
+        """Per-round hop of the (possibly two-level) ring.
+
+        Two-level: every round hops the intra-node ring; at each phase
+        start a snapshot of the current tensors starts around the
+        inter-node ring into ``buffer_list`` (in flight for a whole
+        phase); at the phase-end round the arrived buffers are swapped in
+        as that round's received data (reference ``comm.py:221-254``).
+        """
+        if not self.double_ring or self.world_size == self.intra_size:
+            return self.send_recv(tensor_list, dest_list)
+        intra = self.intra_size
+        if r % intra == 1 and r // intra != self.inter_size - 1:
+            if not self._buffers_match(self.buffer_list, tensor_list):
+                self.buffer_list = [torch.empty_like(t) for t in tensor_list]
+            send_buffer = [t.clone() for t in tensor_list]
+            self._inter_ops += self._make_ring_ops(
+                send_buffer, self.buffer_list, self.local_group2
+            )
+        if r % intra == 0 and r != 0:
+            # the inter transfer is due: its buffers become this round's
+            # received data (no intra hop this round)
+            for i in range(len(dest_list)):
+                dest_list[i], self.buffer_list[i] = (
+                    self.buffer_list[i],
+                    dest_list[i],
+                )
+            self._wait_inter = True
+        else:
+            self.send_recv(tensor_list, dest_list, self.local_group)
 
     def double_ring_send_recv_q(self, tensor_list, dest_list, r=0):
-        self.send_recv(tensor_list, dest_list)
+        """Travelling-dq hop (reference ``comm.py:187-218``): intra hops
+        carry dq along with its q; at each phase boundary the accumulated
+        dq is merged into the inter-ring buffer and sent onward, and the
+        local accumulation restarts from zero; the final call (r = W+1)
+        flushes the inter buffer and takes the last intra hop."""
+        if not self.double_ring or self.world_size == self.intra_size:
+            return self.send_recv(tensor_list, dest_list)
+        intra = self.intra_size
+        if r % intra == 1 and r != 1:
+            if not self._buffers_match(self.buffer_list, tensor_list):
+                self.buffer_list = [torch.empty_like(t) for t in tensor_list]
+                send_buffer = [t.clone() for t in tensor_list]
+                self._inter_ops += self._make_ring_ops(
+                    send_buffer, self.buffer_list, self.local_group2
+                )
+            else:
+                self.wait(True)
+                add_list = [t + b for t, b in zip(tensor_list, self.buffer_list)]
+                self._inter_ops += self._make_ring_ops(
+                    add_list, self.buffer_list, self.local_group2
+                )
+            if r // intra != self.inter_size:
+                for d in dest_list:
+                    d.zero_()
+            else:  # final flush (r == W + 1)
+                self.commit()
+                self.wait(True)
+                self.send_recv(self.buffer_list, dest_list, self.local_group)
+                self.commit()
+                self.wait()
+        else:
+            self.send_recv(tensor_list, dest_list, self.local_group)
 
     def commit(self):
         if self.ops:
             self.reqs += dist.batch_isend_irecv(self.ops)
             self.ops = []
+        if self._inter_ops:
+            self._inter_reqs += dist.batch_isend_irecv(self._inter_ops)
+            self._inter_ops = []
 
-    def wait(self):
-        for req in self.reqs:
-            req.wait()
-        self.reqs = []
+    def wait(self, force_wait_inter=False):
+        if self._wait_inter or force_wait_inter:
+            for req in self._inter_reqs:
+                req.wait()
+            self._inter_reqs = []
+            self._wait_inter = False
+        else:
+            for req in self.reqs:
+                req.wait()
+            self.reqs = []

@@ -43,12 +43,23 @@ __all__ = ["burst_attn_func", "burst_attn_func_striped", "OpBurstAttn", "OpBurst
 
 
 def get_partition_id(double_group, r):
-    """Which source-rank offset's K/V (fwd) or Q (bwd) is held at round r.
+    """Source index of the K/V (fwd) or Q (bwd) chunk held at round r.
 
-    Single ring: offset r-1 (reference ``burst_attn_interface.py:20-37``;
-    the double-group formula is multi-node-only and lands with the double
-    ring)."""
-    return r - 1
+    Single ring: r-1 (equivalent to "source rank <= my rank" in the
+    causal dispatch).  Double ring: the source chunk's global rank, from
+    the intra/inter positions (reference burst_attn_interface.py:20-37).
+    """
+    if not double_group or double_group[0] is None:
+        return r - 1
+    import torch.distributed as dist
+
+    intra = dist.get_world_size(double_group[0])
+    inter = dist.get_world_size(double_group[1])
+    ii = dist.get_rank(double_group[0])
+    ir = dist.get_rank(double_group[1])
+    return ((ir - (r - 1) // intra) % inter) * intra + (
+        (ii - (r - 1) % intra) % intra
+    )
 
 
 def _record_stream(*tensors):
