@@ -129,3 +129,24 @@ print("OK")
     r = subprocess.run([sys.executable, "-c", code], capture_output=True,
                        text=True, cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
     assert r.returncode == 0 and "OK" in r.stdout, r.stdout + r.stderr
+
+
+@pytest.mark.parametrize("causal,striped", [(True, False), (True, True)])
+def test_ring_odd_world_size(causal, striped):
+    """W=3: odd ring size exercises the even/odd P2P ordering with two
+    adjacent even ranks and non-power-of-two chunking."""
+    _PORT[0] += 1
+    ctx = mp.get_context("spawn")
+    fail_q = ctx.SimpleQueue()
+    try:
+        mp.spawn(
+            _worker,
+            args=(3, _PORT[0], causal, striped, False, False, fail_q),
+            nprocs=3,
+            join=True,
+        )
+    except Exception:
+        msgs = []
+        while not fail_q.empty():
+            msgs.append(fail_q.get())
+        raise AssertionError("odd-ring test failed:\n" + "\n".join(msgs))
