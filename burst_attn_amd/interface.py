@@ -164,8 +164,10 @@ class OpBurstAttn(torch.autograd.Function):
         comm_bufs = [torch.empty_like(k), torch.empty_like(v)]
         half = q.shape[1] // 2
         state = None  # provider-owned carry-in accumulator (in-kernel merge)
+        record = []
         for r in range(1, W + 1):
             offset = get_partition_id(double_group, r)
+            record.append(offset)
             split_kv = offset <= rank  # kv origin precedes this rank's chunks
             if r != W:
                 ring.double_ring_send_recv([k, v], comm_bufs, r)
@@ -185,6 +187,7 @@ class OpBurstAttn(torch.autograd.Function):
                 kv, comm_bufs = _record_stream(*comm_bufs), [k, v]
                 k, v = kv
                 ring.wait()
+        _logger.info("fwd record of rank %d: %s", get_rank(), record)
         return _finalize_fwd(ctx, P, q, ori_k, ori_v, state)
 
     @staticmethod

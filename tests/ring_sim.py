@@ -77,7 +77,6 @@ def simulate_ring(P, q_full, k_full, v_full, do_full, W, scale, causal,
             q, do, lse = qs[src], dos[src], lses[src]
             delta = P.bwd_preprocess(os_[src].to(q.dtype), do)
             if striped:
-                causal_shift = (r - 1) <= rank and r != 1
                 # striped bwd: q from an EARLIER rank is shifted
                 causal_shift = ((r - 1) <= rank) and r != 1
                 if not causal_shift or not causal:
