@@ -29,7 +29,6 @@
 
 namespace {
 
-constexpr int NTHREADS = 512;  // 8 waves
 
 // OUT_STATE=0: write normalised o (fp32) + lse — the stateless tile.
 // OUT_STATE=1: carry-in/carry-out accumulator state (acc = unnormalised
@@ -41,8 +40,11 @@ constexpr int NTHREADS = 512;  // 8 waves
 //         1 = V row-major (tr16-swizzled) + ds_read_tr16_b64 fragments
 // SUBT: 0 = joint softmax over the 64-kv tile; 1 = per-32-subtile online
 // updates (lets subtile-0 PV MFMAs overlap subtile-1 QK/softmax)
-template <typename T, int D, int KVBLK, int OUT_STATE, int VPATH, int SUBT>
-__global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
+// NT: threads per workgroup (512 = 8 waves x 1 block/CU;
+//     256 = 4 waves x 2 blocks/CU — decoupled barrier groups)
+template <typename T, int D, int KVBLK, int OUT_STATE, int VPATH, int SUBT,
+          int NT = 512>
+__global__ __launch_bounds__(NT) void attn_fwd_kernel(
     const T* __restrict__ q, const T* __restrict__ k, const T* __restrict__ v,
     float* __restrict__ o, float* __restrict__ lse,
     int Sq, int Sk, int N,
@@ -59,7 +61,8 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
   using frag = typename MT::frag;
   constexpr int SWZ_K = (D == 128) ? 15 : 7;  // K image rows are 2*D bytes
   constexpr int SWZ_V = 7;                    // V^T image rows are 128 bytes
-  constexpr int PT = (KVBLK * D / 8) / NTHREADS;
+  constexpr int PT = (KVBLK * D / 8) / NT;
+  constexpr int QROWS = NT / 2;  // q rows per workgroup (32 per wave)
   static_assert(PT >= 1, "tile must fill at least one chunk per thread");
 
   // single LDS object: [2 buffers][K row-major | V transposed][KVBLK*D]
@@ -76,7 +79,7 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
   const int hi = lane >> 5;
   const int n = blockIdx.y;
   const int b = blockIdx.z;
-  const int qb = blockIdx.x * 256 + wave * 32;
+  const int qb = blockIdx.x * QROWS + wave * 32;
   const int q_row = qb + l31;
 
   const T* qp = q + (int64_t)b * q_sb + (int64_t)n * q_sh;
@@ -113,14 +116,15 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
       for (int r = 0; r < 16; ++r) ot[dt][r] = arow[dt * 32 + ba_crow(r, hi)];
   }
 
-  const int kv_limit = causal ? min(Sk, (int)(blockIdx.x + 1) * 256) : Sk;
+  const int kv_limit =
+      causal ? min(Sk, (int)(blockIdx.x + 1) * QROWS) : Sk;
   const int nt = (kv_limit + KVBLK - 1) / KVBLK;
 
   auto issue_loads = [&](int tile, u32x4_t* kreg, u32x4_t* vreg) {
     const int kv0 = tile * KVBLK;
 #pragma unroll
     for (int c = 0; c < PT; ++c) {
-      const int flat = tid + c * NTHREADS;
+      const int flat = tid + c * NT;
       const int row = flat / (D / 8);
       const int col8 = flat % (D / 8);
       const int kvg = kv0 + row;
@@ -137,7 +141,7 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
   auto write_lds = [&](int buf, const u32x4_t* kreg, const u32x4_t* vreg) {
 #pragma unroll
     for (int c = 0; c < PT; ++c) {
-      const int flat = tid + c * NTHREADS;
+      const int flat = tid + c * NT;
       const int row = flat / (D / 8);
       const int col8 = flat % (D / 8);
       const int byte = ba_swz<SWZ_K>(row * (2 * D) + col8 * 16, row);
@@ -157,7 +161,7 @@ __global__ __launch_bounds__(NTHREADS) void attn_fwd_kernel(
   }
   // static priority for the younger dispatch half (T5 static form):
   // wave-uniform condition via readfirstlane, one s_setprio, no flips
-  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+  if (NT == 512 && __builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
     __builtin_amdgcn_s_setprio(1);
 
   int cur = 0;
@@ -442,17 +446,25 @@ static int launch_fwd(const void* q, const void* k, const void* v, float* o,
     const char* e = getenv("BA_FWD_SUBT");
     return e ? atoi(e) : 1;  // per-subtile softmax pipeline (+3% measured)
   }();
-  dim3 grid((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B);
-#define FWD_LAUNCH(VP, ST)                                                    \
-  attn_fwd_kernel<T, D, 64, 0, VP, ST>                                        \
-      <<<grid, dim3(NTHREADS), 0, (hipStream_t)stream>>>(                     \
+  static const int ntw = [] {
+    const char* e = getenv("BA_FWD_NT");
+    return e ? atoi(e) : 512;
+  }();
+#define FWD_LAUNCH(VP, ST, NTV)                                               \
+  attn_fwd_kernel<T, D, 64, 0, VP, ST, NTV>                                   \
+      <<<dim3((unsigned)((Sq + NTV / 2 - 1) / (NTV / 2)), (unsigned)N,        \
+              (unsigned)B),                                                   \
+         dim3(NTV), 0, (hipStream_t)stream>>>(                                \
           (const T*)q, (const T*)k, (const T*)v, o, lse, (int)Sq, (int)Sk,    \
           (int)N, qs[0], qs[1], qs[2], ks[0], ks[1], ks[2], vs[0], vs[1],     \
           vs[2], scale, causal, nullptr, nullptr, nullptr, 0, 0, 0, 0, 0, 0)
-  if (vpath == 0 && subt == 0) FWD_LAUNCH(0, 0);
-  else if (vpath == 0 && subt == 1) FWD_LAUNCH(0, 1);
-  else if (vpath == 1 && subt == 0) FWD_LAUNCH(1, 0);
-  else FWD_LAUNCH(1, 1);
+  if (ntw == 256) {
+    if (vpath == 0) FWD_LAUNCH(0, 1, 256);
+    else FWD_LAUNCH(1, 1, 256);
+  } else if (vpath == 0 && subt == 0) FWD_LAUNCH(0, 0, 512);
+  else if (vpath == 0 && subt == 1) FWD_LAUNCH(0, 1, 512);
+  else if (vpath == 1 && subt == 0) FWD_LAUNCH(1, 0, 512);
+  else FWD_LAUNCH(1, 1, 512);
 #undef FWD_LAUNCH
   BA_CHECK_LAUNCH();
   return 0;
@@ -470,22 +482,30 @@ static int launch_fwd_accum(const void* q, const void* k, const void* v,
     const char* e = getenv("BA_FWD_VPATH");
     return e ? atoi(e) : 0;
   }();
-  dim3 grid((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B);
   static const int subt = [] {
     const char* e = getenv("BA_FWD_SUBT");
     return e ? atoi(e) : 1;  // per-subtile softmax pipeline (+3% measured)
   }();
-#define FWD_ALAUNCH(VP, ST)                                                   \
-  attn_fwd_kernel<T, D, 64, 1, VP, ST>                                        \
-      <<<grid, dim3(NTHREADS), 0, (hipStream_t)stream>>>(                     \
+  static const int ntw = [] {
+    const char* e = getenv("BA_FWD_NT");
+    return e ? atoi(e) : 512;
+  }();
+#define FWD_ALAUNCH(VP, ST, NTV)                                              \
+  attn_fwd_kernel<T, D, 64, 1, VP, ST, NTV>                                   \
+      <<<dim3((unsigned)((Sq + NTV / 2 - 1) / (NTV / 2)), (unsigned)N,        \
+              (unsigned)B),                                                   \
+         dim3(NTV), 0, (hipStream_t)stream>>>(                                \
           (const T*)q, (const T*)k, (const T*)v, nullptr, nullptr, (int)Sq,   \
           (int)Sk, (int)N, qs[0], qs[1], qs[2], ks[0], ks[1], ks[2], vs[0],   \
           vs[1], vs[2], scale, causal, acc, m, l, as[0], as[1], as[2],        \
           mls[0], mls[1], carry_in)
-  if (vpath == 0 && subt == 0) FWD_ALAUNCH(0, 0);
-  else if (vpath == 0 && subt == 1) FWD_ALAUNCH(0, 1);
-  else if (vpath == 1 && subt == 0) FWD_ALAUNCH(1, 0);
-  else FWD_ALAUNCH(1, 1);
+  if (ntw == 256) {
+    if (vpath == 0) FWD_ALAUNCH(0, 1, 256);
+    else FWD_ALAUNCH(1, 1, 256);
+  } else if (vpath == 0 && subt == 0) FWD_ALAUNCH(0, 0, 512);
+  else if (vpath == 0 && subt == 1) FWD_ALAUNCH(0, 1, 512);
+  else if (vpath == 1 && subt == 0) FWD_ALAUNCH(1, 0, 512);
+  else FWD_ALAUNCH(1, 1, 512);
 #undef FWD_ALAUNCH
   BA_CHECK_LAUNCH();
   return 0;
