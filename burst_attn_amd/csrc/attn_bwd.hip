@@ -226,8 +226,7 @@ __global__ __launch_bounds__(512) void bwd_dq_kernel(
 // backward executes 8 tile GEMMs total vs flash-attn's 5).
 // MODE: 0 = dV (dv^T += mfma(dO^T, P)); 1 = dK (dk^T += mfma(Q^T, dS));
 //       2 = fused dK+dV (both accumulators; 8-wave; may spill — A/B)
-// QBLK: streamed q-tile rows; dV has register headroom for 128 (fewer
-//       barriers/staging passes per GEMM), dK is at the 256-VGPR cap.
+// QBLK: streamed q-tile rows (128 for dV measured -8%: 27-VGPR spills).
 template <typename T, int D, int MODE, int QBLK = 64>
 __global__ __launch_bounds__(512) void bwd_dkv_kernel(
     const T* __restrict__ dout, const T* __restrict__ q,
@@ -524,7 +523,7 @@ static int launch_bwd(const void* dout, const void* q, const void* k,
       scale, causal);
   BA_CHECK_LAUNCH();
   dim3 grid_kv((unsigned)((Sk + 255) / 256), (unsigned)N, (unsigned)B);
-  bwd_dkv_kernel<T, D, 0, 128><<<grid_kv, 512, 0, (hipStream_t)stream>>>(
+  bwd_dkv_kernel<T, D, 0, 64><<<grid_kv, 512, 0, (hipStream_t)stream>>>(
       (const T*)dout, (const T*)q, (const T*)k, (const T*)v, delta, lse, dv,
       nullptr, (int)Sq, (int)Sk, (int)N, gs[0], gs[1], gs[2], qs[0], qs[1], qs[2],
       ks[0], ks[1], ks[2], vs[0], vs[1], vs[2], ds[0], ds[1], ls[0], ls[1],
