@@ -390,8 +390,37 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
     if (active) {
 #pragma unroll
       for (int qs = 0; qs < QBLK / 32; ++qs) {
-        f32x16_t st = (f32x16_t)(0.f);
+        // ---- augmentation fold: the per-row constants ride the MFMA.
+        // One extra k-slice per GEMM with B = [1,1,0...] (lanes of the
+        // low half) and A carrying a row constant split hi/lo into two
+        // fp16/bf16 elements (the split keeps the exponent-domain error
+        // at fp32 roundoff): S accumulates -lse*log2e/c2 so the exp
+        // argument is just st*c2; dP accumulates -delta so dS is just
+        // p*dpt*scale.  Replaces 32 per-row LDS reads + fma/sub chains
+        // (68% WAIT_ANY and 24 spilled VGPRs on the dK kernel).
+        const float2 ld2 = *(const float2*)&ldsF(cur)[2 * (qs * 32 + l31)];
+        frag ones01, ones0, a_lse, a_dlt;
+        {
+          u32x4_t z = {0, 0, 0, 0};
+          ones01 = __builtin_bit_cast(frag, z);
+          a_lse = __builtin_bit_cast(frag, z);
+          a_dlt = __builtin_bit_cast(frag, z);
+          if (hi == 0) {
+            const float av = -ld2.x / c2;
+            const T ah = (T)av;
+            a_lse[0] = ah;
+            a_lse[1] = (T)(av - (float)ah);
+            const T dh = (T)(-ld2.y);
+            a_dlt[0] = dh;
+            a_dlt[1] = (T)(-ld2.y - (float)dh);
+            ones01[0] = (T)1.f;
+            ones01[1] = (T)1.f;
+          }
+          ones0 = ones01;
+        }
+        f32x16_t st = MT::mma(a_lse, ones01, (f32x16_t)(0.f));
         f32x16_t dpt = (f32x16_t)(0.f);
+        if (MODE >= 1) dpt = MT::mma(a_dlt, ones0, dpt);
 #pragma unroll
         for (int s = 0; s < D / 16; ++s) {
           frag qfr = ba_ld_rowslice<T, D, SWZ>(ldsQ(cur), qs * 32 + l31,
@@ -403,35 +432,21 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
             dpt = MT::mma(gfr, vf[s], dpt);
           }
         }
-        // {lse2, delta} reads batched 8 rows ahead of use: one LDS wait
-        // per 8 instead of a serial load-use chain (68% WAIT_ANY on dK)
 #pragma unroll
-        for (int rb = 0; rb < 2; ++rb) {
-          float2 ld2v[8];
-#pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            const int q_loc = qs * 32 + ba_crow(rb * 8 + j, 0) + 4 * hi;
-            ld2v[j] = *(const float2*)&ldsF(cur)[2 * q_loc];
-          }
-#pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            const int r = rb * 8 + j;
-            const int q_loc = qs * 32 + ba_crow(r, 0) + 4 * hi;
-            const int q_g = q0 + q_loc;
-            const bool valid =
-                q_g < Sq && kv_col < Sk && (!causal || q_g >= kv_col);
-            const float2 ld2 = ld2v[j];
-            const float e =
-                valid ? __builtin_fmaf(st[r], c2, -ld2.x) : BA_NEG_BIG;
-            const float p = ba_exp2(e);
-            if (MODE == 0) {
-              st[r] = p;  // P for dV
-            } else if (MODE == 1) {
-              st[r] = p * (dpt[r] - ld2.y) * scale;  // dS for dK
-            } else {
-              dpt[r] = p * (dpt[r] - ld2.y) * scale;  // dS
-              st[r] = p;                              // P
-            }
+        for (int r = 0; r < 16; ++r) {
+          const int q_loc = qs * 32 + ba_crow(r, 0) + 4 * hi;
+          const int q_g = q0 + q_loc;
+          const bool valid =
+              q_g < Sq && kv_col < Sk && (!causal || q_g >= kv_col);
+          const float e = valid ? st[r] * c2 : BA_NEG_BIG;
+          const float p = ba_exp2(e);
+          if (MODE == 0) {
+            st[r] = p;  // P for dV
+          } else if (MODE == 1) {
+            st[r] = p * dpt[r] * scale;  // dS for dK
+          } else {
+            dpt[r] = p * dpt[r] * scale;  // dS
+            st[r] = p;                    // P
           }
         }
         frag f01[2];
