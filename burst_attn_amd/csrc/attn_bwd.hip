@@ -418,8 +418,9 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
           }
           ones0 = ones01;
         }
-        f32x16_t st = (f32x16_t)(0.f);
+        f32x16_t st = MT::mma(a_lse, ones01, (f32x16_t)(0.f));
         f32x16_t dpt = (f32x16_t)(0.f);
+        if (MODE >= 1) dpt = MT::mma(a_dlt, ones0, dpt);
 #pragma unroll
         for (int s = 0; s < D / 16; ++s) {
           frag qfr = ba_ld_rowslice<T, D, SWZ>(ldsQ(cur), qs * 32 + l31,
@@ -431,10 +432,6 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
             dpt = MT::mma(gfr, vf[s], dpt);
           }
         }
-        // aug MFMAs LAST: their ld2-load/convert latency hides under the
-        // s-loop instead of heading the accumulator dependency chain
-        st = MT::mma(a_lse, ones01, st);
-        if (MODE >= 1) dpt = MT::mma(a_dlt, ones0, dpt);
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           const int q_loc = qs * 32 + ba_crow(r, 0) + 4 * hi;
