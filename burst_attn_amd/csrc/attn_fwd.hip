@@ -42,8 +42,13 @@ namespace {
 // updates (lets subtile-0 PV MFMAs overlap subtile-1 QK/softmax)
 // NT: threads per workgroup (512 = 8 waves x 1 block/CU;
 //     256 = 4 waves x 2 blocks/CU — decoupled barrier groups)
+// NBUF: LDS tile buffers. 2 = stage one tile ahead, barrier every tile.
+// 4 = stage two tiles ahead, barrier every OTHER tile (a buffer is
+// rewritten two tiles after its last read, so one barrier in any two
+// consecutive tile boundaries separates every write from its readers
+// and every reader from the overwrite).
 template <typename T, int D, int KVBLK, int OUT_STATE, int VPATH, int SUBT,
-          int NT = 512>
+          int NT = 512, int NBUF = 2>
 __global__ __launch_bounds__(NT) void attn_fwd_kernel(
     const T* __restrict__ q, const T* __restrict__ k, const T* __restrict__ v,
     float* __restrict__ o, float* __restrict__ lse,
@@ -65,8 +70,8 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
   constexpr int QROWS = NT / 2;  // q rows per workgroup (32 per wave)
   static_assert(PT >= 1, "tile must fill at least one chunk per thread");
 
-  // single LDS object: [2 buffers][K row-major | V transposed][KVBLK*D]
-  __shared__ T lds[2 * 2 * KVBLK * D];
+  // single LDS object: [NBUF buffers][K row-major | V transposed][KVBLK*D]
+  __shared__ T lds[NBUF * 2 * KVBLK * D];
   auto ldsK = [&](int buf) -> T* { return lds + buf * (2 * KVBLK * D); };
   auto ldsVT = [&](int buf) -> T* {
     return lds + buf * (2 * KVBLK * D) + KVBLK * D;
@@ -153,10 +158,14 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
     }
   };
 
-  {  // prologue: tile 0
+  {  // prologue: tiles 0..NBUF-2
     u32x4_t kreg[PT], vreg[PT];
     issue_loads(0, kreg, vreg);
     write_lds(0, kreg, vreg);
+    if (NBUF == 4 && nt > 1) {
+      issue_loads(1, kreg, vreg);
+      write_lds(1, kreg, vreg);
+    }
     __syncthreads();
   }
   // static priority for the younger dispatch half (T5 static form):
@@ -164,12 +173,13 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
   if (NT == 512 && __builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
     __builtin_amdgcn_s_setprio(1);
 
-  int cur = 0;
+  constexpr int AHEAD = NBUF == 4 ? 2 : 1;
   for (int t = 0; t < nt; ++t) {
     const int kv0 = t * KVBLK;
-    const bool has_next = (t + 1) < nt;
+    const int cur = t % NBUF;
+    const bool has_next = (t + AHEAD) < nt;
     u32x4_t kreg[PT], vreg[PT];
-    if (has_next) issue_loads(t + 1, kreg, vreg);
+    if (has_next) issue_loads(t + AHEAD, kreg, vreg);
 
     const bool active = !causal || (kv0 <= qb + 31);
     if (active && SUBT == 1) {
@@ -324,9 +334,8 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
       }
     }
 
-    if (has_next) write_lds(cur ^ 1, kreg, vreg);
-    __syncthreads();
-    cur ^= 1;
+    if (has_next) write_lds((t + AHEAD) % NBUF, kreg, vreg);
+    if (NBUF == 2 || (t & 1) || t + 1 >= nt) __syncthreads();
   }
 
   // ---- epilogue
@@ -450,6 +459,30 @@ static int launch_fwd(const void* q, const void* k, const void* v, float* o,
     const char* e = getenv("BA_FWD_NT");
     return e ? atoi(e) : 512;
   }();
+  static const int nbuf = [] {
+    const char* e = getenv("BA_FWD_NBUF");
+    return e ? atoi(e) : 2;
+  }();
+  if (nbuf == 4 && ntw == 512) {
+    if (vpath == 0)
+      attn_fwd_kernel<T, D, 64, 0, 0, 1, 512, 4>
+          <<<dim3((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B),
+             dim3(512), 0, (hipStream_t)stream>>>(
+              (const T*)q, (const T*)k, (const T*)v, o, lse, (int)Sq,
+              (int)Sk, (int)N, qs[0], qs[1], qs[2], ks[0], ks[1], ks[2],
+              vs[0], vs[1], vs[2], scale, causal, nullptr, nullptr, nullptr,
+              0, 0, 0, 0, 0, 0);
+    else
+      attn_fwd_kernel<T, D, 64, 0, 1, 1, 512, 4>
+          <<<dim3((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B),
+             dim3(512), 0, (hipStream_t)stream>>>(
+              (const T*)q, (const T*)k, (const T*)v, o, lse, (int)Sq,
+              (int)Sk, (int)N, qs[0], qs[1], qs[2], ks[0], ks[1], ks[2],
+              vs[0], vs[1], vs[2], scale, causal, nullptr, nullptr, nullptr,
+              0, 0, 0, 0, 0, 0);
+    BA_CHECK_LAUNCH();
+    return 0;
+  }
 #define FWD_LAUNCH(VP, ST, NTV)                                               \
   attn_fwd_kernel<T, D, 64, 0, VP, ST, NTV>                                   \
       <<<dim3((unsigned)((Sq + NTV / 2 - 1) / (NTV / 2)), (unsigned)N,        \
@@ -490,6 +523,29 @@ static int launch_fwd_accum(const void* q, const void* k, const void* v,
     const char* e = getenv("BA_FWD_NT");
     return e ? atoi(e) : 512;
   }();
+  static const int nbuf = [] {
+    const char* e = getenv("BA_FWD_NBUF");
+    return e ? atoi(e) : 2;
+  }();
+  if (nbuf == 4 && ntw == 512) {
+    dim3 grid4((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B);
+    if (vpath == 0)
+      attn_fwd_kernel<T, D, 64, 1, 0, 1, 512, 4>
+          <<<grid4, dim3(512), 0, (hipStream_t)stream>>>(
+              (const T*)q, (const T*)k, (const T*)v, nullptr, nullptr,
+              (int)Sq, (int)Sk, (int)N, qs[0], qs[1], qs[2], ks[0], ks[1],
+              ks[2], vs[0], vs[1], vs[2], scale, causal, acc, m, l, as[0],
+              as[1], as[2], mls[0], mls[1], carry_in);
+    else
+      attn_fwd_kernel<T, D, 64, 1, 1, 1, 512, 4>
+          <<<grid4, dim3(512), 0, (hipStream_t)stream>>>(
+              (const T*)q, (const T*)k, (const T*)v, nullptr, nullptr,
+              (int)Sq, (int)Sk, (int)N, qs[0], qs[1], qs[2], ks[0], ks[1],
+              ks[2], vs[0], vs[1], vs[2], scale, causal, acc, m, l, as[0],
+              as[1], as[2], mls[0], mls[1], carry_in);
+    BA_CHECK_LAUNCH();
+    return 0;
+  }
 #define FWD_ALAUNCH(VP, ST, NTV)                                              \
   attn_fwd_kernel<T, D, 64, 1, VP, ST, NTV>                                   \
       <<<dim3((unsigned)((Sq + NTV / 2 - 1) / (NTV / 2)), (unsigned)N,        \
