@@ -48,6 +48,8 @@ def parse_args():
     p.add_argument("--causal", action="store_true")
     p.add_argument("--optimize-bwd-comm", action="store_true",
                    help="ring the fp32 delta instead of o in backward")
+    p.add_argument("--striped", action="store_true",
+                   help="use the striped causal variant (OpBurstAttnStrip)")
     p.add_argument("--no-cpu-baseline", action="store_true")
     p.add_argument("--no-bwd", action="store_true",
                    help="skip the fwd+bwd timing leg")
@@ -143,7 +145,7 @@ def main():
     dtype = torch.float16 if args.dtype == "fp16" else torch.bfloat16
     dev = torch.device("cuda")
 
-    from burst_attn_amd import burst_attn_func
+    from burst_attn_amd import burst_attn_func, burst_attn_func_striped
     from burst_attn_amd._ext import load_extension
 
     ext = load_extension()
@@ -158,17 +160,17 @@ def main():
     log(f"[bench] rank {rank}/{world} s_local={s_local} dtype={args.dtype}")
 
     causal = args.causal
+    attn = burst_attn_func_striped if args.striped else burst_attn_func
 
     def fwd_step():
         with torch.no_grad():
-            burst_attn_func(q, k, v, None, "cuda", causal)
+            attn(q, k, v, None, "cuda", causal)
 
     def fwdbwd_step():
         qg = q.detach().requires_grad_()
         kg = k.detach().requires_grad_()
         vg = v.detach().requires_grad_()
-        o = burst_attn_func(qg, kg, vg, None, "cuda", causal,
-                            args.optimize_bwd_comm)
+        o = attn(qg, kg, vg, None, "cuda", causal, args.optimize_bwd_comm)
         torch.autograd.grad(o, (qg, kg, vg), do)
 
     t_fwd = time_loop(fwd_step, args.steps, args.warmup)
@@ -234,7 +236,8 @@ def main():
             "config": {
                 "workload": f"b={args.batch} seq={args.seq} h={args.heads} "
                             f"d={args.dim} {'causal' if causal else 'non-causal'}",
-                "parallelism": f"ring sequence parallel, sp{world}",
+                "parallelism": f"ring sequence parallel, sp{world}"
+                               + (" striped" if args.striped else ""),
             },
             "fwd_bwd_TFLOPs_per_GPU": round(tflops_fb, 2) if tflops_fb else None,
             "fwd_bwd_ms_per_step": round(t_fb * 1e3, 2) if t_fb else None,
