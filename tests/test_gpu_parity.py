@@ -393,3 +393,31 @@ def test_error_paths():
 
     with pytest.raises(ValueError):
         burst_attn_func(q, q, q, None, "math")
+
+
+@pytest.mark.timeout(600)
+def test_large_size_direct_parity():
+    """Direct (non-property) parity at s=8192 — two orders of magnitude
+    above the per-case tile tests; the oracle runs blockwise on CPU in
+    ~tens of seconds."""
+    b, s, n, d = 1, 8192, 2, 128
+    dtype = torch.float16
+    q = _rand(b, s, n, d, dtype, 71)
+    k = _rand(b, s, n, d, dtype, 72)
+    v = _rand(b, s, n, d, dtype, 73)
+    do = _rand(b, s, n, d, dtype, 74)
+    scale = 1.0 / math.sqrt(d)
+    ext = _ext()
+    o, lse = ext.attn_fwd(q, k, v, scale, True)
+    delta = ext.attn_bwd_preprocess(o.to(dtype), do)
+    dq, dk, dv = ext.attn_bwd(do, q, k, v, delta, lse, scale, True, False)
+    o_ref, lse_ref = oracle.tile_fwd(q.cpu(), k.cpu(), v.cpu(), scale, True,
+                                     q_block=1024, k_block=1024)
+    torch.testing.assert_close(o.cpu(), o_ref, **TOL[dtype])
+    torch.testing.assert_close(lse.cpu(), lse_ref, rtol=1e-3, atol=2e-2)
+    dq_r, dk_r, dv_r = oracle.tile_bwd(do.cpu(), q.cpu(), k.cpu(), v.cpu(),
+                                       lse_ref, scale, True, o=o_ref,
+                                       q_block=1024, k_block=1024)
+    torch.testing.assert_close(dv.cpu(), dv_r, **BWD_TOL[dtype])
+    torch.testing.assert_close(dk.cpu(), dk_r, **BWD_TOL[dtype])
+    torch.testing.assert_close(dq.cpu(), dq_r, **BWD_TOL[dtype])
