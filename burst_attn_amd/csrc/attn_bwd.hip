@@ -126,15 +126,12 @@ __global__ __launch_bounds__(512) void bwd_dq_kernel(
     for (int c = 0; c < PT; ++c) {
       const int flat = tid + c * NT;
       const int row = flat / (D / 8), col8 = flat % (D / 8);
+      // clamped instead of guarded (see attn_fwd.hip): out-of-range kv
+      // rows re-load row Sk-1; their S values are masked to p = 0
       const int kvg = kv0 + row;
-      if (kvg < Sk) {
-        kreg[c] = *(const u32x4_t*)(kp + (int64_t)kvg * k_ss + col8 * 8);
-        vreg[c] = *(const u32x4_t*)(vp + (int64_t)kvg * v_ss + col8 * 8);
-      } else {
-        u32x4_t z = {0, 0, 0, 0};
-        kreg[c] = z;
-        vreg[c] = z;
-      }
+      const int kvc = kvg < Sk ? kvg : (Sk - 1);
+      kreg[c] = *(const u32x4_t*)(kp + (int64_t)kvc * k_ss + col8 * 8);
+      vreg[c] = *(const u32x4_t*)(vp + (int64_t)kvc * v_ss + col8 * 8);
     }
   };
   auto write_lds = [&](int buf, const u32x4_t* kreg, const u32x4_t* vreg) {
@@ -324,15 +321,13 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
     for (int c = 0; c < PT; ++c) {
       const int flat = tid + c * NT;
       const int row = flat / (D / 8), col8 = flat % (D / 8);
+      // clamped instead of guarded (see attn_fwd.hip): out-of-range q
+      // rows re-load row Sq-1; the q_g < Sq mask zeroes their P/dS, so
+      // they contribute garbage x 0 to the dV/dK contractions
       const int qg = q0 + row;
-      if (qg < Sq) {
-        qreg[c] = *(const u32x4_t*)(qp + (int64_t)qg * q_ss + col8 * 8);
-        greg[c] = *(const u32x4_t*)(gp + (int64_t)qg * g_ss + col8 * 8);
-      } else {
-        u32x4_t z = {0, 0, 0, 0};
-        qreg[c] = z;
-        greg[c] = z;
-      }
+      const int qc = qg < Sq ? qg : (Sq - 1);
+      qreg[c] = *(const u32x4_t*)(qp + (int64_t)qc * q_ss + col8 * 8);
+      greg[c] = *(const u32x4_t*)(gp + (int64_t)qc * g_ss + col8 * 8);
     }
     // lse*log2e and delta rows, interleaved {lse2, delta} per q row so
     // the compute loop reads one 8B word per row

@@ -132,15 +132,14 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
       const int flat = tid + c * NT;
       const int row = flat / (D / 8);
       const int col8 = flat % (D / 8);
+      // clamped index instead of a guarded load: the guard compiles to
+      // an exec-mask branch around every load pair (serialising the
+      // staging); clamped rows re-load row Sk-1, whose values are
+      // nullified by the kv-range mask (p = 0) in the softmax
       const int kvg = kv0 + row;
-      if (kvg < Sk) {
-        kreg[c] = *(const u32x4_t*)(kp + (int64_t)kvg * k_ss + col8 * 8);
-        vreg[c] = *(const u32x4_t*)(vp + (int64_t)kvg * v_ss + col8 * 8);
-      } else {
-        u32x4_t z = {0, 0, 0, 0};
-        kreg[c] = z;
-        vreg[c] = z;
-      }
+      const int kvc = kvg < Sk ? kvg : (Sk - 1);
+      kreg[c] = *(const u32x4_t*)(kp + (int64_t)kvc * k_ss + col8 * 8);
+      vreg[c] = *(const u32x4_t*)(vp + (int64_t)kvc * v_ss + col8 * 8);
     }
   };
   auto write_lds = [&](int buf, const u32x4_t* kreg, const u32x4_t* vreg) {
