@@ -112,6 +112,19 @@ class Ring:
         if self.double_ring:
             self.intra_size = dist.get_world_size(self.local_group)
             self.inter_size = dist.get_world_size(self.local_group2)
+            # the phase-boundary staging assumes a proper 2-level split;
+            # intra_size == 1 or a non-divisible split would swap against a
+            # never-populated buffer_list (IndexError deep in a ring round)
+            assert self.intra_size > 1, (
+                "double_ring needs intra_size > 1 (pass double_group="
+                "[None, None] for a flat ring)"
+            )
+            assert self.world_size % self.intra_size == 0, (
+                "double_ring needs world_size divisible by intra_size"
+            )
+            assert self.intra_size * self.inter_size == self.world_size, (
+                "double_ring groups must tile the world: intra * inter == W"
+            )
         else:
             self.intra_size = self.world_size
             self.inter_size = 1

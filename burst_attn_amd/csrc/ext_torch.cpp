@@ -38,6 +38,23 @@ void fill_strides(const at::Tensor& t, int64_t s[3]) {
   s[2] = t.stride(2);
 }
 
+// cross-tensor consistency vs q: batch/heads/head_dim agree, k/v seqlens
+// agree, dtypes are equal — a mismatch must raise here, not read out of
+// bounds on device.
+void check_qkv_consistent(const at::Tensor& q, const at::Tensor& k,
+                          const at::Tensor& v) {
+  TORCH_CHECK(k.size(0) == q.size(0) && v.size(0) == q.size(0),
+              "batch mismatch");
+  TORCH_CHECK(k.size(2) == q.size(2) && v.size(2) == q.size(2),
+              "heads mismatch");
+  TORCH_CHECK(k.size(3) == q.size(3) && v.size(3) == q.size(3),
+              "head_dim mismatch");
+  TORCH_CHECK(v.size(1) == k.size(1), "k/v seqlen mismatch");
+  TORCH_CHECK(q.scalar_type() == k.scalar_type() &&
+                  q.scalar_type() == v.scalar_type(),
+              "q/k/v dtype mismatch");
+}
+
 #define BA_CALL(expr)                                                      \
   do {                                                                     \
     int rc_ = (expr);                                                      \
@@ -53,13 +70,7 @@ std::vector<at::Tensor> attn_fwd(const at::Tensor& q, const at::Tensor& k,
   check_qkv(v, "v");
   const auto B = q.size(0), Sq = q.size(1), N = q.size(2), D = q.size(3);
   const auto Sk = k.size(1);
-  TORCH_CHECK(k.size(0) == B && v.size(0) == B, "batch mismatch");
-  TORCH_CHECK(k.size(2) == N && v.size(2) == N, "heads mismatch");
-  TORCH_CHECK(k.size(3) == D && v.size(3) == D, "head_dim mismatch");
-  TORCH_CHECK(v.size(1) == Sk, "k/v seqlen mismatch");
-  TORCH_CHECK(q.scalar_type() == k.scalar_type() &&
-                  q.scalar_type() == v.scalar_type(),
-              "q/k/v dtype mismatch");
+  check_qkv_consistent(q, k, v);
   auto o = at::empty({B, Sq, N, D}, q.options().dtype(at::kFloat));
   auto lse = at::empty({B, N, Sq}, q.options().dtype(at::kFloat));
   int64_t qs[3], ks[3], vs[3];
@@ -83,6 +94,7 @@ void attn_fwd_accum(const at::Tensor& q, const at::Tensor& k,
   check_qkv(v, "v");
   const auto B = q.size(0), Sq = q.size(1), N = q.size(2), D = q.size(3);
   const auto Sk = k.size(1);
+  check_qkv_consistent(q, k, v);
   TORCH_CHECK(acc.scalar_type() == at::kFloat && m.scalar_type() == at::kFloat
                   && l.scalar_type() == at::kFloat, "state must be fp32");
   TORCH_CHECK(acc.size(1) == Sq && m.size(2) == Sq && l.size(2) == Sq,
@@ -149,6 +161,11 @@ std::vector<at::Tensor> attn_bwd(const at::Tensor& dout, const at::Tensor& q,
   check_qkv(v, "v");
   const auto B = q.size(0), Sq = q.size(1), N = q.size(2), D = q.size(3);
   const auto Sk = k.size(1);
+  check_qkv_consistent(q, k, v);
+  TORCH_CHECK(dout.size(0) == B && dout.size(1) == Sq && dout.size(2) == N &&
+                  dout.size(3) == D,
+              "dout/q shape mismatch");
+  TORCH_CHECK(dout.scalar_type() == q.scalar_type(), "dout/q dtype mismatch");
   TORCH_CHECK(delta.scalar_type() == at::kFloat && lse.scalar_type() == at::kFloat,
               "delta/lse must be fp32");
   TORCH_CHECK(delta.dim() == 3 && lse.dim() == 3, "delta/lse must be [B,N,S]");

@@ -162,6 +162,11 @@ class OpBurstAttn(torch.autograd.Function):
         W, rank = ring.world_size, ring.rank
         ori_k, ori_v = replicate(k), replicate(v)
         comm_bufs = [torch.empty_like(k), torch.empty_like(v)]
+        if causal:
+            assert q.shape[1] % 2 == 0, (
+                "zigzag causal needs an even per-rank seqlen (the rank holds "
+                "two half-chunks)"
+            )
         half = q.shape[1] // 2
         state = None  # provider-owned carry-in accumulator (in-kernel merge)
         record = []
@@ -200,6 +205,11 @@ class OpBurstAttn(torch.autograd.Function):
         ring = Ring(group, double_group)
         dq_ring = Ring(group, ctx.dq_group if ctx.dq_group is not None else double_group)
         W, rank = ring.world_size, ring.rank
+        if ctx.causal:
+            assert q.shape[1] % 2 == 0, (
+                "zigzag causal needs an even per-rank seqlen (the rank holds "
+                "two half-chunks)"
+            )
         half = q.shape[1] // 2
 
         dq = torch.zeros(q.shape, dtype=torch.float32, device=q.device)
