@@ -7,21 +7,39 @@
 //     dv += p^T do;   dp = do v^T;   ds = p * (dp - delta) * scale
 //     dq += ds k;     dk += ds^T q
 //
-// Split into three kernels (two-pass dq: deterministic by construction, no
-// atomics — the `deterministic` flag is accepted and always honoured):
-//   1. bwd_preprocess: delta = rowsum(o * do) fp32   (flash's preprocess)
-//   2. dq kernel  (8 waves, q-block resident, streams k/v):
-//        S^T = mfma(K,Q); dP^T = mfma(V,dO); dS^T in-lane;
-//        dQ^T += mfma(K^T, dS^T)            — all softmax state lane-local
-//   3. dkdv kernel (4 waves, kv-block resident, streams q/do):
-//        S = mfma(Q,K^T); dP = mfma(dO,V^T); dS in-lane;
-//        dV^T += mfma(dO^T, P); dK^T += mfma(Q^T, dS)
+// dq/dk/dv are fp32 ACCUMULATORS: every kernel adds its tile contribution
+// in place (strided), which lets the ring layer accumulate across rounds
+// with no elementwise-add passes or per-round allocations.
+//
+// Two kernel plans behind bahip_attn_bwd:
+//   deterministic=1 (atomic-free, 8 tile GEMMs — flash-attn executes 5):
+//     1. bwd_preprocess: delta = rowsum(o * do) fp32  (flash's preprocess)
+//     2. dq kernel  (8 waves, q-block resident, streams k/v):
+//          S^T = mfma(K,Q); dP^T = mfma(V,dO); dS^T in-lane;
+//          dQ^T += mfma(K^T, dS^T)          — all softmax state lane-local
+//     3. dkdv kernels (kv-block resident, streams q/do):
+//          MODE 0 (dV): S = mfma(Q,K^T); dV^T += mfma(dO^T, P)
+//          MODE 1 (dK): + dP = mfma(dO,V^T); dK^T += mfma(Q^T, dS)
+//   deterministic=0 (default, 6 tile GEMMs — the flash-attn plan):
+//     1. bwd_preprocess
+//     2. bwd_dkq kernel (kv-resident, FLIPPED orientation: S^T = mfma(K,Q)
+//        with kv on the MFMA row axis and q on the lane axis, so lse/delta
+//        are lane-local — no per-row constant folding, no serial LDS->MFMA
+//        prologue):  S^T, dP^T, then
+//          dK^T += mfma(Q^T, dS)   (dS fragments via a per-wave LDS
+//                                   transpose of the dS^T D-layout)
+//          dQ   += mfma(dS^T, K^T) (fragments built in-register; partials
+//                                   LDS-reduced across waves, flushed with
+//                                   one global fp32 atomic-add per
+//                                   workgroup q-subtile)
+//     3. dkdv MODE 0 (dV) as above.
 // Operand scheme and LDS swizzle: see attn_common.h.
 
 #include "attn_common.h"
 #include "../../include/burst_attn_hip.h"
 
 #include <stdio.h>
+#include <stdlib.h>
 
 namespace {
 
@@ -66,6 +84,7 @@ __global__ __launch_bounds__(512) void bwd_dq_kernel(
     int64_t k_sb, int64_t k_ss, int64_t k_sh,
     int64_t v_sb, int64_t v_ss, int64_t v_sh,
     int64_t d_sb, int64_t d_sh, int64_t l_sb, int64_t l_sh,
+    int64_t dq_sb, int64_t dq_ss, int64_t dq_sh,
     float scale, int causal) {
   using MT = mfma_traits<T>;
   using frag = typename MT::frag;
@@ -206,12 +225,14 @@ __global__ __launch_bounds__(512) void bwd_dq_kernel(
   }
 
   if (q_row < Sq) {
-    float* drow = dq + (((int64_t)b * Sq + q_row) * N + n) * D;
+    // accumulate epilogue: each (b, q_row, n) row is written by exactly
+    // one lane, so a plain read-add-write stays deterministic
+    float* drow = dq + b * dq_sb + (int64_t)q_row * dq_ss + n * dq_sh;
 #pragma unroll
     for (int dt = 0; dt < D / 32; ++dt)
 #pragma unroll
       for (int r = 0; r < 16; ++r)
-        drow[dt * 32 + ba_crow(r, hi)] = dqt[dt][r];
+        drow[dt * 32 + ba_crow(r, hi)] += dqt[dt][r];
   }
 }
 
@@ -236,6 +257,7 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
     int64_t k_sb, int64_t k_ss, int64_t k_sh,
     int64_t v_sb, int64_t v_ss, int64_t v_sh,
     int64_t d_sb, int64_t d_sh, int64_t l_sb, int64_t l_sh,
+    int64_t o_sb, int64_t o_ss, int64_t o_sh,
     float scale, int causal) {
   using MT = mfma_traits<T>;
   using frag = typename MT::frag;
@@ -476,20 +498,302 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
   }
 
   if (kv_col < Sk) {
-    float* row = dout_acc + (((int64_t)b * Sk + kv_col) * N + n) * D;
+    // accumulate epilogue (single-writer per row: read-add-write)
+    float* row = dout_acc + b * o_sb + (int64_t)kv_col * o_ss + n * o_sh;
 #pragma unroll
     for (int dt = 0; dt < D / 32; ++dt)
 #pragma unroll
       for (int r = 0; r < 16; ++r)
-        row[dt * 32 + ba_crow(r, hi)] = acc[dt][r];
+        row[dt * 32 + ba_crow(r, hi)] += acc[dt][r];
     if (MODE == 2) {
-      float* row2 = dout_acc2 + (((int64_t)b * Sk + kv_col) * N + n) * D;
+      float* row2 = dout_acc2 + b * o_sb + (int64_t)kv_col * o_ss + n * o_sh;
 #pragma unroll
       for (int dt = 0; dt < D / 32; ++dt)
 #pragma unroll
         for (int r = 0; r < 16; ++r)
-          row2[dt * 32 + ba_crow(r, hi)] = acc2[dt][r];
+          row2[dt * 32 + ba_crow(r, hi)] += acc2[dt][r];
     }
+  }
+}
+
+// ============== fused dK + dQ kernel (kv-resident, flipped) ============
+// FLIPPED orientation: S^T = mfma(K, Q) puts kv on the MFMA row axis and
+// q on the lane axis, so lse/delta are LANE-LOCAL scalars (read straight
+// from global, coalesced) — the split dK kernel's per-q-row constants
+// (the "augmentation fold" MFMAs and their serial LDS->MFMA prologue,
+// 65% SQ_WAIT in the round-1 PMCs) disappear.
+//   dS^T sits in D-layout [kv rows][q cols].  From it:
+//   * dQ fragments [row=q][contr=kv] come from ba_build_frag_pair
+//     IN-REGISTER; dQ += mfma(dS^T, K^T) against a persistent K^T LDS
+//     image (K is kv-block resident, staged transposed once).  Per-wave
+//     partials (each wave owns 32 kv) are LDS-reduced (ds_add) across the
+//     workgroup's 256 kv and flushed with ONE global fp32 atomic-add per
+//     q-subtile — flash-attn's accumulation plan (hence deterministic=0
+//     only).
+//   * dK fragments [row=kv][contr=q] need the other orientation: a
+//     per-wave LDS transpose (16 b16 scatter writes + 2 b128 reads, no
+//     barrier — wave-private scratch); dK^T += mfma(Q^T, dS) stays
+//     register-resident exactly like the split kernel.
+// FUSE_DQ=0 degenerates to a flipped dK-only kernel (atomic-free — legal
+// for deterministic=1; kept for A/B against the split MODE 1).
+template <typename T, int D, int FUSE_DQ>
+__global__ __launch_bounds__(512) void bwd_dkq_kernel(
+    const T* __restrict__ dout, const T* __restrict__ q,
+    const T* __restrict__ k, const T* __restrict__ v,
+    const float* __restrict__ delta, const float* __restrict__ lse,
+    float* __restrict__ dq, float* __restrict__ dk,
+    int Sq, int Sk, int N,
+    int64_t g_sb, int64_t g_ss, int64_t g_sh,
+    int64_t q_sb, int64_t q_ss, int64_t q_sh,
+    int64_t k_sb, int64_t k_ss, int64_t k_sh,
+    int64_t v_sb, int64_t v_ss, int64_t v_sh,
+    int64_t d_sb, int64_t d_sh, int64_t l_sb, int64_t l_sh,
+    int64_t dq_sb, int64_t dq_ss, int64_t dq_sh,
+    int64_t dk_sb, int64_t dk_ss, int64_t dk_sh,
+    float scale, int causal) {
+  using MT = mfma_traits<T>;
+  using frag = typename MT::frag;
+  constexpr int QBLK = 32;    // streamed q rows per tile (1 subtile)
+  constexpr int KVWG = 256;   // kv columns per workgroup (8 waves x 32)
+  constexpr int NT = 512;
+  constexpr int SWZ = (D == 128) ? 15 : 7;  // Q/dO row-major images
+  constexpr int SWZ_QT = 3;                 // Q^T image rows are 64 B
+  constexpr int SCRW = 64;  // scratch row width (elems; 32 used) -> 128 B
+  constexpr int SWZ_SC = 7;                 // conflict-free dsf reads
+  constexpr int CHUNKS = QBLK * (D / 8);    // staged 8-elem chunks per img
+
+  // one LDS object (a second __shared__ forces vmcnt(0) per ds_read):
+  // [2 buf x {Q row-major | dO row-major | Q^T}] [K^T persistent]
+  // [8 x per-wave dS scratch] [fp32 dq reduce buffer]
+  constexpr int STREAM = 2 * 3 * QBLK * D;
+  constexpr int KT_E = FUSE_DQ ? D * KVWG : 0;
+  constexpr int SCR_E = 8 * 32 * SCRW;
+  constexpr int REDU_E = FUSE_DQ ? QBLK * D * (int)(4 / sizeof(T)) : 0;
+  __shared__ T lds[STREAM + KT_E + SCR_E + REDU_E];
+  auto ldsQ = [&](int buf) -> T* { return lds + buf * (3 * QBLK * D); };
+  auto ldsG = [&](int buf) -> T* {
+    return lds + buf * (3 * QBLK * D) + QBLK * D;
+  };
+  auto ldsQT = [&](int buf) -> T* {
+    return lds + buf * (3 * QBLK * D) + 2 * QBLK * D;
+  };
+  T* ldsKT = lds + STREAM;
+  auto ldsSC = [&](int w) -> T* { return lds + STREAM + KT_E + w * 32 * SCRW; };
+  float* redu = (float*)(lds + STREAM + KT_E + SCR_E);
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int l31 = lane & 31, hi = lane >> 5;
+  const int n = blockIdx.y, b = blockIdx.z;
+  const int kvb = blockIdx.x * KVWG + wave * 32;  // wave's kv block
+  const int kv_col = kvb + l31;
+
+  const T* qp = q + b * q_sb + (int64_t)n * q_sh;
+  const T* gp = dout + b * g_sb + (int64_t)n * g_sh;
+  const T* kp = k + b * k_sb + (int64_t)n * k_sh;
+  const T* vp = v + b * v_sb + (int64_t)n * v_sh;
+  const float* dp_ = delta + b * d_sb + n * d_sh;
+  const float* lp_ = lse + b * l_sb + n * l_sh;
+
+  // resident K and V fragments [row=kv][contr=d] (A operands of S^T/dP^T)
+  frag kf[D / 16], vf[D / 16];
+#pragma unroll
+  for (int s = 0; s < D / 16; ++s) {
+    if (kv_col < Sk) {
+      kf[s] = __builtin_bit_cast(
+          frag, *(const u32x4_t*)(kp + (int64_t)kv_col * k_ss + 16 * s + 8 * hi));
+      vf[s] = __builtin_bit_cast(
+          frag, *(const u32x4_t*)(vp + (int64_t)kv_col * v_ss + 16 * s + 8 * hi));
+    } else {
+      u32x4_t z = {0, 0, 0, 0};
+      kf[s] = __builtin_bit_cast(frag, z);
+      vf[s] = __builtin_bit_cast(frag, z);
+    }
+  }
+  const float c2 = scale * BA_LOG2E;
+
+  f32x16_t acc[D / 32];  // dK^T accumulator [d rows][kv col = lane]
+#pragma unroll
+  for (int dt = 0; dt < D / 32; ++dt) acc[dt] = (f32x16_t)(0.f);
+
+  // one-time: zero the reduce buffer; stage K^T [D rows][KVWG] transposed
+  if (FUSE_DQ) {
+    for (int i = tid; i < QBLK * D; i += NT) redu[i] = 0.f;
+    constexpr int PTK = (KVWG * D / 8) / NT;
+#pragma unroll
+    for (int c = 0; c < PTK; ++c) {
+      const int flat = tid + c * NT;
+      const int kv = flat / (D / 8), col8 = flat % (D / 8);
+      const int kvg = blockIdx.x * KVWG + kv;
+      const int kvc = kvg < Sk ? kvg : (Sk - 1);
+      u32x4_t ch = *(const u32x4_t*)(kp + (int64_t)kvc * k_ss + col8 * 8);
+      ba_st_transposed<T, KVWG, 15, 0>(ldsKT, kv, col8 * 8, ch);
+    }
+  }
+
+  const int t0 = causal ? (blockIdx.x * KVWG) / QBLK : 0;
+  const int nt = (Sq + QBLK - 1) / QBLK;
+
+  // per-tile prefetch: Q and dO chunks + this lane's lse*log2e and delta
+  auto issue_loads = [&](int tile, u32x4_t* qreg, u32x4_t* greg, float* lse2,
+                         float* dlt) {
+    const int q0 = tile * QBLK;
+    constexpr int PT = (CHUNKS + NT - 1) / NT;
+#pragma unroll
+    for (int c = 0; c < PT; ++c) {
+      const int flat = tid + c * NT;
+      if (CHUNKS % NT == 0 || flat < CHUNKS) {
+        const int row = flat / (D / 8), col8 = flat % (D / 8);
+        const int qg = q0 + row;
+        const int qc = qg < Sq ? qg : (Sq - 1);  // clamped (see attn_fwd.hip)
+        qreg[c] = *(const u32x4_t*)(qp + (int64_t)qc * q_ss + col8 * 8);
+        greg[c] = *(const u32x4_t*)(gp + (int64_t)qc * g_ss + col8 * 8);
+      }
+    }
+    const int qg = q0 + l31;
+    const int qc = qg < Sq ? qg : (Sq - 1);
+    *lse2 = lp_[qc] * BA_LOG2E;  // lane-local; invalid rows masked later
+    *dlt = dp_[qc];
+  };
+  auto write_lds = [&](int buf, const u32x4_t* qreg, const u32x4_t* greg) {
+    constexpr int PT = (CHUNKS + NT - 1) / NT;
+#pragma unroll
+    for (int c = 0; c < PT; ++c) {
+      const int flat = tid + c * NT;
+      if (CHUNKS % NT == 0 || flat < CHUNKS) {
+        const int row = flat / (D / 8), col8 = flat % (D / 8);
+        const int byte = ba_swz<SWZ>(row * (2 * D) + col8 * 16, row);
+        *(u32x4_t*)((char*)ldsQ(buf) + byte) = qreg[c];
+        *(u32x4_t*)((char*)ldsG(buf) + byte) = greg[c];
+        ba_st_transposed<T, QBLK, SWZ_QT, 3>(ldsQT(buf), row, col8 * 8,
+                                             qreg[c]);
+      }
+    }
+  };
+
+  float lse2_c, dlt_c;
+  {
+    u32x4_t qreg[(CHUNKS + NT - 1) / NT], greg[(CHUNKS + NT - 1) / NT];
+    issue_loads(t0, qreg, greg, &lse2_c, &dlt_c);
+    write_lds(0, qreg, greg);
+    __syncthreads();
+  }
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);  // T5 static form (younger half)
+
+  int cur = 0;
+  for (int t = t0; t < nt; ++t) {
+    const int q0 = t * QBLK;
+    const bool has_next = (t + 1) < nt;
+    u32x4_t qreg[(CHUNKS + NT - 1) / NT], greg[(CHUNKS + NT - 1) / NT];
+    float lse2_n = 0.f, dlt_n = 0.f;
+    if (has_next) issue_loads(t + 1, qreg, greg, &lse2_n, &dlt_n);
+
+    // waves whose kv block is wholly after this q tile are masked anyway
+    const bool active = !causal || (q0 + QBLK - 1 >= kvb);
+    if (active) {
+      // ---- S^T = mfma(K, Q), dP^T = mfma(V, dO): [kv rows][q cols]
+      f32x16_t st = (f32x16_t)(0.f), dpt = (f32x16_t)(0.f);
+#pragma unroll
+      for (int s = 0; s < D / 16; ++s) {
+        frag qfr = ba_ld_rowslice<T, D, SWZ>(ldsQ(cur), l31, 16 * s + 8 * hi);
+        frag gfr = ba_ld_rowslice<T, D, SWZ>(ldsG(cur), l31, 16 * s + 8 * hi);
+        st = MT::mma(kf[s], qfr, st);
+        dpt = MT::mma(vf[s], gfr, dpt);
+      }
+      // ---- dS^T = p * (dP^T - delta) * scale, lse/delta LANE-LOCAL
+      const int q_g = q0 + l31;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int kv_g = kvb + ba_crow(r, 0) + 4 * hi;
+        const bool valid =
+            q_g < Sq && kv_g < Sk && (!causal || q_g >= kv_g);
+        const float e = valid ? __builtin_fmaf(st[r], c2, -lse2_c) : BA_NEG_BIG;
+        const float p = ba_exp2(e);
+        st[r] = p * (dpt[r] - dlt_c) * scale;
+      }
+      // ---- per-wave scratch transpose: dS^T D-layout -> dS fragments
+      {
+        T* sc = ldsSC(wave);
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int row = ba_crow(r, 0) + 4 * hi;  // kv-local
+          const int byte = ba_swz<SWZ_SC, 0>(row * (2 * SCRW) + 2 * l31, row);
+          *(T*)((char*)sc + byte) = (T)st[r];
+        }
+      }
+      if (FUSE_DQ) {
+        // ---- dQ: fragments [row=q][contr=kv] in-register; one dt tile
+        // at a time to keep the partial's register span at 16
+        frag dsT[2];
+        ba_build_frag_pair<T>(st, dsT);
+#pragma unroll
+        for (int dt = 0; dt < D / 32; ++dt) {
+          f32x16_t dqp = (f32x16_t)(0.f);
+#pragma unroll
+          for (int u = 0; u < 2; ++u) {
+            frag ktf = ba_ld_rowslice<T, KVWG, 15, 0>(
+                ldsKT, dt * 32 + l31, wave * 32 + 16 * u + 8 * hi);
+            dqp = MT::mma(dsT[u], ktf, dqp);
+          }
+#pragma unroll
+          for (int r = 0; r < 16; ++r) {
+            const int q_loc = ba_crow(r, 0) + 4 * hi;
+            atomicAdd(&redu[q_loc * D + dt * 32 + l31], dqp[r]);  // ds_add
+          }
+        }
+      }
+      // ---- dK^T += mfma(Q^T, dS)
+      frag dsf[2];
+      {
+        const T* sc = ldsSC(wave);
+        dsf[0] = ba_ld_rowslice<T, SCRW, SWZ_SC, 0>(sc, l31, 8 * hi);
+        dsf[1] = ba_ld_rowslice<T, SCRW, SWZ_SC, 0>(sc, l31, 16 + 8 * hi);
+      }
+#pragma unroll
+      for (int dt = 0; dt < D / 32; ++dt) {
+#pragma unroll
+        for (int u = 0; u < 2; ++u) {
+          frag tf = ba_ld_rowslice<T, QBLK, SWZ_QT, 3>(
+              ldsQT(cur), dt * 32 + l31, 16 * u + 8 * hi);
+          acc[dt] = MT::mma(tf, dsf[u], acc[dt]);
+        }
+      }
+    }
+    if (has_next) write_lds(cur ^ 1, qreg, greg);
+    __syncthreads();
+    if (FUSE_DQ) {
+      // flush the reduced dq partial: ONE atomic per element per
+      // workgroup (the ds_adds above summed the 8 waves' 32-kv partials)
+      const int base = tid * (QBLK * D / NT);  // 8 consecutive floats
+      const int q_loc = base / D, d0 = base % D;
+      const int q_g2 = q0 + q_loc;
+      if (q_g2 < Sq) {
+        float* gdst = dq + b * dq_sb + (int64_t)q_g2 * dq_ss + n * dq_sh + d0;
+#pragma unroll
+        for (int j = 0; j < QBLK * D / NT; ++j) {
+          unsafeAtomicAdd(&gdst[j], redu[base + j]);
+          redu[base + j] = 0.f;
+        }
+      } else {
+#pragma unroll
+        for (int j = 0; j < QBLK * D / NT; ++j) redu[base + j] = 0.f;
+      }
+      __syncthreads();
+    }
+    cur ^= 1;
+    lse2_c = lse2_n;
+    dlt_c = dlt_n;
+  }
+
+  if (kv_col < Sk) {
+    float* row = dk + b * dk_sb + (int64_t)kv_col * dk_ss + n * dk_sh;
+#pragma unroll
+    for (int dt = 0; dt < D / 32; ++dt)
+#pragma unroll
+      for (int r = 0; r < 16; ++r)
+        row[dt * 32 + ba_crow(r, hi)] += acc[dt][r];
   }
 }
 
@@ -517,34 +821,70 @@ extern "C" int bahip_attn_bwd_preprocess(
   return 0;
 }
 
+// plan selection: deterministic=0 -> fused dK+dQ (6 tile GEMMs, atomic
+// dq) + dV; deterministic=1 -> atomic-free split 3-kernel plan.
+// BA_BWD_FUSED=0/1 forces the plan (A/B knob; 1 with deterministic=1 is
+// refused — the fused dq accumulation order is not fixed).
+// BA_BWD_DK_FLIP=1 swaps the split plan's MODE-1 dK kernel for the
+// flipped (FUSE_DQ=0) kernel — atomic-free, so deterministic-legal.
 template <typename T, int D>
 static int launch_bwd(const void* dout, const void* q, const void* k,
                       const void* v, const float* delta, const float* lse,
                       float* dq, float* dk, float* dv, int64_t B, int64_t Sq,
                       int64_t Sk, int64_t N, const int64_t* gs,
                       const int64_t* qs, const int64_t* ks, const int64_t* vs,
-                      const int64_t* ds, const int64_t* ls, float scale,
-                      int causal, void* stream) {
-  dim3 grid_dq((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B);
-  bwd_dq_kernel<T, D><<<grid_dq, 512, 0, (hipStream_t)stream>>>(
-      (const T*)dout, (const T*)q, (const T*)k, (const T*)v, delta, lse, dq,
-      (int)Sq, (int)Sk, (int)N, gs[0], gs[1], gs[2], qs[0], qs[1], qs[2],
-      ks[0], ks[1], ks[2], vs[0], vs[1], vs[2], ds[0], ds[1], ls[0], ls[1],
-      scale, causal);
-  BA_CHECK_LAUNCH();
+                      const int64_t* ds, const int64_t* ls,
+                      const int64_t* dqs, const int64_t* dks,
+                      const int64_t* dvs, float scale, int causal,
+                      int deterministic, void* stream) {
+  static const int env_fused = [] {
+    const char* e = getenv("BA_BWD_FUSED");
+    return e ? atoi(e) : -1;
+  }();
+  static const int dk_flip = [] {
+    const char* e = getenv("BA_BWD_DK_FLIP");
+    return e ? atoi(e) : 0;
+  }();
+  const bool fused = env_fused >= 0 ? (env_fused && !deterministic)
+                                    : !deterministic;
   dim3 grid_kv((unsigned)((Sk + 255) / 256), (unsigned)N, (unsigned)B);
+  if (!fused) {
+    dim3 grid_dq((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B);
+    bwd_dq_kernel<T, D><<<grid_dq, 512, 0, (hipStream_t)stream>>>(
+        (const T*)dout, (const T*)q, (const T*)k, (const T*)v, delta, lse, dq,
+        (int)Sq, (int)Sk, (int)N, gs[0], gs[1], gs[2], qs[0], qs[1], qs[2],
+        ks[0], ks[1], ks[2], vs[0], vs[1], vs[2], ds[0], ds[1], ls[0], ls[1],
+        dqs[0], dqs[1], dqs[2], scale, causal);
+    BA_CHECK_LAUNCH();
+  }
   bwd_dkv_kernel<T, D, 0, 64><<<grid_kv, 512, 0, (hipStream_t)stream>>>(
       (const T*)dout, (const T*)q, (const T*)k, (const T*)v, delta, lse, dv,
-      nullptr, (int)Sq, (int)Sk, (int)N, gs[0], gs[1], gs[2], qs[0], qs[1], qs[2],
-      ks[0], ks[1], ks[2], vs[0], vs[1], vs[2], ds[0], ds[1], ls[0], ls[1],
-      scale, causal);
+      nullptr, (int)Sq, (int)Sk, (int)N, gs[0], gs[1], gs[2], qs[0], qs[1],
+      qs[2], ks[0], ks[1], ks[2], vs[0], vs[1], vs[2], ds[0], ds[1], ls[0],
+      ls[1], dvs[0], dvs[1], dvs[2], scale, causal);
   BA_CHECK_LAUNCH();
-  bwd_dkv_kernel<T, D, 1><<<grid_kv, 512, 0, (hipStream_t)stream>>>(
-      (const T*)dout, (const T*)q, (const T*)k, (const T*)v, delta, lse, dk,
-      nullptr, (int)Sq, (int)Sk, (int)N, gs[0], gs[1], gs[2], qs[0], qs[1], qs[2],
-      ks[0], ks[1], ks[2], vs[0], vs[1], vs[2], ds[0], ds[1], ls[0], ls[1],
-      scale, causal);
-  BA_CHECK_LAUNCH();
+  if (fused) {
+    bwd_dkq_kernel<T, D, 1><<<grid_kv, 512, 0, (hipStream_t)stream>>>(
+        (const T*)dout, (const T*)q, (const T*)k, (const T*)v, delta, lse,
+        dq, dk, (int)Sq, (int)Sk, (int)N, gs[0], gs[1], gs[2], qs[0], qs[1],
+        qs[2], ks[0], ks[1], ks[2], vs[0], vs[1], vs[2], ds[0], ds[1], ls[0],
+        ls[1], dqs[0], dqs[1], dqs[2], dks[0], dks[1], dks[2], scale, causal);
+    BA_CHECK_LAUNCH();
+  } else if (dk_flip) {
+    bwd_dkq_kernel<T, D, 0><<<grid_kv, 512, 0, (hipStream_t)stream>>>(
+        (const T*)dout, (const T*)q, (const T*)k, (const T*)v, delta, lse,
+        dq, dk, (int)Sq, (int)Sk, (int)N, gs[0], gs[1], gs[2], qs[0], qs[1],
+        qs[2], ks[0], ks[1], ks[2], vs[0], vs[1], vs[2], ds[0], ds[1], ls[0],
+        ls[1], dqs[0], dqs[1], dqs[2], dks[0], dks[1], dks[2], scale, causal);
+    BA_CHECK_LAUNCH();
+  } else {
+    bwd_dkv_kernel<T, D, 1><<<grid_kv, 512, 0, (hipStream_t)stream>>>(
+        (const T*)dout, (const T*)q, (const T*)k, (const T*)v, delta, lse,
+        dk, nullptr, (int)Sq, (int)Sk, (int)N, gs[0], gs[1], gs[2], qs[0],
+        qs[1], qs[2], ks[0], ks[1], ks[2], vs[0], vs[1], vs[2], ds[0], ds[1],
+        ls[0], ls[1], dks[0], dks[1], dks[2], scale, causal);
+    BA_CHECK_LAUNCH();
+  }
   return 0;
 }
 
@@ -555,31 +895,19 @@ extern "C" int bahip_attn_bwd(
     const int64_t do_strides[3], const int64_t q_strides[3],
     const int64_t k_strides[3], const int64_t v_strides[3],
     const int64_t delta_strides[2], const int64_t lse_strides[2],
-    float softmax_scale, int causal, int deterministic, int dtype,
-    void* stream) {
-  (void)deterministic;  // two-pass dq is deterministic by construction
+    const int64_t dq_strides[3], const int64_t dk_strides[3],
+    const int64_t dv_strides[3], float softmax_scale, int causal,
+    int deterministic, int dtype, void* stream) {
   if (causal && Sq != Sk) return 1001;
-  if (D == 128 && dtype == BAHIP_BF16)
-    return launch_bwd<__bf16, 128>(dout, q, k, v, delta, lse, dq, dk, dv, B,
-                                   Sq, Sk, N, do_strides, q_strides, k_strides,
-                                   v_strides, delta_strides, lse_strides,
-                                   softmax_scale, causal, stream);
-  if (D == 128 && dtype == BAHIP_F16)
-    return launch_bwd<_Float16, 128>(dout, q, k, v, delta, lse, dq, dk, dv, B,
-                                     Sq, Sk, N, do_strides, q_strides,
-                                     k_strides, v_strides, delta_strides,
-                                     lse_strides, softmax_scale, causal,
-                                     stream);
-  if (D == 64 && dtype == BAHIP_BF16)
-    return launch_bwd<__bf16, 64>(dout, q, k, v, delta, lse, dq, dk, dv, B, Sq,
-                                  Sk, N, do_strides, q_strides, k_strides,
-                                  v_strides, delta_strides, lse_strides,
-                                  softmax_scale, causal, stream);
-  if (D == 64 && dtype == BAHIP_F16)
-    return launch_bwd<_Float16, 64>(dout, q, k, v, delta, lse, dq, dk, dv, B,
-                                    Sq, Sk, N, do_strides, q_strides,
-                                    k_strides, v_strides, delta_strides,
-                                    lse_strides, softmax_scale, causal,
-                                    stream);
+#define LAUNCH_BWD(T, DD)                                                     \
+  launch_bwd<T, DD>(dout, q, k, v, delta, lse, dq, dk, dv, B, Sq, Sk, N,      \
+                    do_strides, q_strides, k_strides, v_strides,              \
+                    delta_strides, lse_strides, dq_strides, dk_strides,       \
+                    dv_strides, softmax_scale, causal, deterministic, stream)
+  if (D == 128 && dtype == BAHIP_BF16) return LAUNCH_BWD(__bf16, 128);
+  if (D == 128 && dtype == BAHIP_F16) return LAUNCH_BWD(_Float16, 128);
+  if (D == 64 && dtype == BAHIP_BF16) return LAUNCH_BWD(__bf16, 64);
+  if (D == 64 && dtype == BAHIP_F16) return LAUNCH_BWD(_Float16, 64);
+#undef LAUNCH_BWD
   return 1002;
 }

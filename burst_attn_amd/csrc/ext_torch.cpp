@@ -133,12 +133,21 @@ std::vector<at::Tensor> attn_fwd_finalize(const at::Tensor& acc,
   return {o, lse};
 }
 
-at::Tensor attn_bwd_preprocess(const at::Tensor& o, const at::Tensor& dout) {
+at::Tensor attn_bwd_preprocess(const at::Tensor& o, const at::Tensor& dout,
+                               c10::optional<at::Tensor> out) {
   check_qkv(o, "o");
   check_qkv(dout, "dout");
   const auto B = o.size(0), S = o.size(1), N = o.size(2), D = o.size(3);
   TORCH_CHECK(dout.sizes() == o.sizes(), "o/dout shape mismatch");
-  auto delta = at::empty({B, N, S}, o.options().dtype(at::kFloat));
+  at::Tensor delta;
+  if (out.has_value()) {
+    delta = *out;
+    TORCH_CHECK(delta.scalar_type() == at::kFloat && delta.is_contiguous() &&
+                    delta.sizes() == at::IntArrayRef({B, N, S}),
+                "preprocess out must be contiguous fp32 [B,N,S]");
+  } else {
+    delta = at::empty({B, N, S}, o.options().dtype(at::kFloat));
+  }
   int64_t os[3], gs[3];
   fill_strides(o, os);
   fill_strides(dout, gs);
@@ -150,11 +159,24 @@ at::Tensor attn_bwd_preprocess(const at::Tensor& o, const at::Tensor& dout) {
   return delta;
 }
 
-std::vector<at::Tensor> attn_bwd(const at::Tensor& dout, const at::Tensor& q,
-                                 const at::Tensor& k, const at::Tensor& v,
-                                 const at::Tensor& delta,
-                                 const at::Tensor& lse, double softmax_scale,
-                                 bool causal, bool deterministic) {
+void check_grad_out(const at::Tensor& g, const at::Tensor& like,
+                    const char* name) {
+  TORCH_CHECK(g.is_cuda(), name, " must be a GPU tensor");
+  TORCH_CHECK(g.scalar_type() == at::kFloat, name,
+              " accumulator must be fp32");
+  TORCH_CHECK(g.dim() == 4, name, " must be [B,S,N,D]");
+  TORCH_CHECK(g.stride(3) == 1, name, " head_dim must be contiguous");
+  TORCH_CHECK(g.sizes() == like.sizes(), name, " shape mismatch");
+}
+
+// in-place accumulate: dq/dk/dv (fp32, strided views allowed) += tile
+// contribution.  The ring layer's round accumulation happens HERE, not
+// in python adds (reference burst_attn_interface.py:379-390).
+void attn_bwd_accum(const at::Tensor& dout, const at::Tensor& q,
+                    const at::Tensor& k, const at::Tensor& v,
+                    const at::Tensor& delta, const at::Tensor& lse,
+                    double softmax_scale, bool causal, bool deterministic,
+                    at::Tensor& dq, at::Tensor& dk, at::Tensor& dv) {
   check_qkv(dout, "dout");
   check_qkv(q, "q");
   check_qkv(k, "k");
@@ -173,14 +195,17 @@ std::vector<at::Tensor> attn_bwd(const at::Tensor& dout, const at::Tensor& q,
               "delta/lse seq dim must be contiguous");
   TORCH_CHECK(delta.size(2) == Sq && lse.size(2) == Sq,
               "delta/lse seqlen mismatch");
-  auto dq = at::empty({B, Sq, N, D}, q.options().dtype(at::kFloat));
-  auto dk = at::empty({B, Sk, N, D}, q.options().dtype(at::kFloat));
-  auto dv = at::empty({B, Sk, N, D}, q.options().dtype(at::kFloat));
-  int64_t gs[3], qs[3], ks[3], vs[3];
+  check_grad_out(dq, q, "dq");
+  check_grad_out(dk, k, "dk");
+  check_grad_out(dv, v, "dv");
+  int64_t gs[3], qs[3], ks[3], vs[3], dqs[3], dks[3], dvs[3];
   fill_strides(dout, gs);
   fill_strides(q, qs);
   fill_strides(k, ks);
   fill_strides(v, vs);
+  fill_strides(dq, dqs);
+  fill_strides(dk, dks);
+  fill_strides(dv, dvs);
   int64_t ds[2] = {delta.stride(0), delta.stride(1)};
   int64_t ls[2] = {lse.stride(0), lse.stride(1)};
   auto stream = at::hip::getCurrentHIPStream().stream();
@@ -188,9 +213,24 @@ std::vector<at::Tensor> attn_bwd(const at::Tensor& dout, const at::Tensor& q,
                          v.data_ptr(), delta.data_ptr<float>(),
                          lse.data_ptr<float>(), dq.data_ptr<float>(),
                          dk.data_ptr<float>(), dv.data_ptr<float>(), B, Sq,
-                         Sk, N, D, gs, qs, ks, vs, ds, ls,
+                         Sk, N, D, gs, qs, ks, vs, ds, ls, dqs, dks, dvs,
                          (float)softmax_scale, causal ? 1 : 0,
                          deterministic ? 1 : 0, dtype_code(q), stream));
+}
+
+std::vector<at::Tensor> attn_bwd(const at::Tensor& dout, const at::Tensor& q,
+                                 const at::Tensor& k, const at::Tensor& v,
+                                 const at::Tensor& delta,
+                                 const at::Tensor& lse, double softmax_scale,
+                                 bool causal, bool deterministic) {
+  const auto B = q.size(0), Sq = q.size(1), N = q.size(2), D = q.size(3);
+  const auto Sk = k.size(1);
+  // zero-filled: the kernels accumulate, so this equals fresh outputs
+  auto dq = at::zeros({B, Sq, N, D}, q.options().dtype(at::kFloat));
+  auto dk = at::zeros({B, Sk, N, D}, q.options().dtype(at::kFloat));
+  auto dv = at::zeros({B, Sk, N, D}, q.options().dtype(at::kFloat));
+  attn_bwd_accum(dout, q, k, v, delta, lse, softmax_scale, causal,
+                 deterministic, dq, dk, dv);
   return {dq, dk, dv};
 }
 
@@ -220,8 +260,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "carry-in accumulator fwd tile (in-kernel LSE merge)");
   m.def("attn_fwd_finalize", &attn_fwd_finalize,
         "o = acc/l (cast), lse from state");
-  m.def("attn_bwd_preprocess", &attn_bwd_preprocess, "delta = rowsum(o*do)");
+  m.def("attn_bwd_preprocess", &attn_bwd_preprocess, "delta = rowsum(o*do)",
+        py::arg("o"), py::arg("dout"), py::arg("out") = py::none());
   m.def("attn_bwd", &attn_bwd, "BurstAttention bwd tile (gfx950)");
+  m.def("attn_bwd_accum", &attn_bwd_accum,
+        "bwd tile accumulating into fp32 dq/dk/dv views");
   m.def("mfma_probe", &mfma_probe, "32x32x16 MFMA layout probe");
   m.def("tr16_probe", &tr16_probe, "ds_read_tr16_b64 semantics probe");
 }

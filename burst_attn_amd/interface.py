@@ -30,6 +30,7 @@ Causal load-balancing layouts (see oracle/partition.py):
 """
 
 import math
+import os
 
 import torch
 
@@ -70,6 +71,22 @@ def _record_stream(*tensors):
             if t.is_cuda:
                 t.record_stream(torch.cuda.current_stream())
     return tensors
+
+
+def _dq_wire_dtype(q, dq_ring):
+    """Wire dtype for the travelling dq on the flat ring.
+
+    The reference rings dq in the input dtype (fp16,
+    ``burst_attn_interface.py:301,393``); we keep the in-round
+    accumulation fp32 and cast at the hop, so the payload matches the
+    reference at better numerics.  ``BA_DQ_WIRE=fp32`` restores the fp32
+    wire.  The double ring merges/zeros the travelling buffer itself
+    (``comm.py:187-218`` semantics), so it keeps the fp32 wire."""
+    if dq_ring.world_size <= 1 or dq_ring.double_ring:
+        return None
+    if os.environ.get("BA_DQ_WIRE", "dtype") == "fp32":
+        return None
+    return q.dtype
 
 
 def _check_flash_arg(flash):
@@ -215,15 +232,29 @@ class OpBurstAttn(torch.autograd.Function):
         dq = torch.zeros(q.shape, dtype=torch.float32, device=q.device)
         dk = torch.zeros(k.shape, dtype=torch.float32, device=k.device)
         dv = torch.zeros(v.shape, dtype=torch.float32, device=v.device)
+        # rank-local staging for rounds whose travelling dq is in flight:
+        # kernels accumulate here during the hop, ONE add folds it in
+        # after the wait (at W=1 the kernels hit dq directly)
+        dq_local = torch.zeros_like(dq) if W > 1 else dq
 
         if ctx.optimize_bwd_comm:
             # ring the tiny fp32 delta instead of o (reference :269-278)
             dlt = P.bwd_preprocess(o, grad_output)  # [B,N,S] fp32
+            delta_buf = None
         else:
             dlt = o  # o travels; delta recomputed per round
+            delta_buf = torch.empty(
+                q.shape[0], q.shape[2], q.shape[1], dtype=torch.float32,
+                device=q.device,
+            )
+        wire_dtype = _dq_wire_dtype(q, dq_ring)
+        if wire_dtype is not None:
+            wire_s = torch.empty(q.shape, dtype=wire_dtype, device=q.device)
+            wire_r = torch.empty_like(wire_s)
 
         read_bufs = [torch.empty_like(t) for t in (dlt, grad_output, q, lse)]
-        dq_buf = [torch.empty_like(dq)]
+        # the fp32 swap buffer is only needed on the fp32-wire path
+        dq_buf = [torch.empty_like(dq)] if wire_dtype is None else []
         for r in range(1, W + 1):
             offset = get_partition_id(double_group, r)
             split_q = offset <= rank  # q origin precedes this rank
@@ -231,29 +262,38 @@ class OpBurstAttn(torch.autograd.Function):
                 ring.double_ring_send_recv([dlt, grad_output, q, lse], read_bufs, r)
                 ring.commit()
             if r != 1:
-                dq_ring.double_ring_send_recv_q([dq], dq_buf, r)
+                if wire_dtype is not None:
+                    wire_s.copy_(dq)
+                    dq_ring.send_recv([wire_s], [wire_r])
+                else:
+                    dq_ring.double_ring_send_recv_q([dq], dq_buf, r)
                 dq_ring.commit()
-            delta = dlt if ctx.optimize_bwd_comm else P.bwd_preprocess(dlt, grad_output)
+            delta = (
+                dlt if ctx.optimize_bwd_comm
+                else P.bwd_preprocess(dlt, grad_output, out=delta_buf)
+            )
+            tgt = dq if r == 1 else dq_local
             if r == 1 or not ctx.causal:
-                dq_i, dk_i, dv_i = P.bwd(
+                P.bwd_accum(
                     grad_output, q, k, v, delta, lse, scale, ctx.causal,
-                    ctx.deterministic,
+                    ctx.deterministic, tgt, dk, dv,
                 )
                 acc = "full"
             elif split_q:
                 # travelling q's second half attends my full kv (:322-345)
-                dq_i, dk_i, dv_i = P.bwd(
+                P.bwd_accum(
                     grad_output[:, half:], q[:, half:], k, v,
                     delta[:, :, half:], lse[:, :, half:], scale, False,
-                    ctx.deterministic,
+                    ctx.deterministic, tgt[:, half:], dk, dv,
                 )
                 acc = "q_half"
             else:
                 # travelling q (all of it) attends only my kv first half
                 # (:347-367)
-                dq_i, dk_i, dv_i = P.bwd(
+                P.bwd_accum(
                     grad_output, q, k[:, :half], v[:, :half], delta, lse,
                     scale, False, ctx.deterministic,
+                    tgt, dk[:, :half], dv[:, :half],
                 )
                 acc = "kv_half"
             if r != W:
@@ -265,25 +305,30 @@ class OpBurstAttn(torch.autograd.Function):
             ring.wait()
             if r != 1:
                 dq_ring.wait()
-                recv, dq_buf = _record_stream(*dq_buf), [dq]
-                dq = recv[0]
-            if acc == "full":
-                dq += dq_i
-                dk += dk_i
-                dv += dv_i
-            elif acc == "q_half":
-                dq[:, half:] += dq_i
-                dk += dk_i
-                dv += dv_i
-            else:
-                dq += dq_i
-                dk[:, :half] += dk_i
-                dv[:, :half] += dv_i
+                if wire_dtype is not None:
+                    dq.copy_(wire_r)
+                else:
+                    recv, dq_buf = _record_stream(*dq_buf), [dq]
+                    dq = recv[0]
+                # fold this round's local contribution into the received dq
+                if acc == "q_half":
+                    dq[:, half:] += dq_local[:, half:]
+                    dq_local[:, half:].zero_()
+                else:
+                    dq += dq_local
+                    dq_local.zero_()
         # one extra hop returns the travelling dq to its owner (:393-396)
-        dq_ring.double_ring_send_recv_q([dq], dq_buf, W + 1)
-        dq_ring.commit()
-        dq_ring.wait()
-        dq = _record_stream(*dq_buf)[0]
+        if wire_dtype is not None:
+            wire_s.copy_(dq)
+            dq_ring.send_recv([wire_s], [wire_r])
+            dq_ring.commit()
+            dq_ring.wait()
+            dq.copy_(wire_r)
+        else:
+            dq_ring.double_ring_send_recv_q([dq], dq_buf, W + 1)
+            dq_ring.commit()
+            dq_ring.wait()
+            dq = _record_stream(*dq_buf)[0]
         return (
             dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype),
             None, None, None, None, None, None, None,
@@ -341,14 +386,25 @@ class OpBurstAttnStrip(torch.autograd.Function):
         dq = torch.zeros(q.shape, dtype=torch.float32, device=q.device)
         dk = torch.zeros(k.shape, dtype=torch.float32, device=k.device)
         dv = torch.zeros(v.shape, dtype=torch.float32, device=v.device)
+        dq_local = torch.zeros_like(dq) if W > 1 else dq
 
         if ctx.optimize_bwd_comm:
             dlt = P.bwd_preprocess(o, grad_output)
+            delta_buf = None
         else:
             dlt = o
+            delta_buf = torch.empty(
+                q.shape[0], q.shape[2], q.shape[1], dtype=torch.float32,
+                device=q.device,
+            )
+        wire_dtype = _dq_wire_dtype(q, dq_ring)
+        if wire_dtype is not None:
+            wire_s = torch.empty(q.shape, dtype=wire_dtype, device=q.device)
+            wire_r = torch.empty_like(wire_s)
 
         read_bufs = [torch.empty_like(t) for t in (dlt, grad_output, q, lse)]
-        dq_buf = [torch.empty_like(dq)]
+        # the fp32 swap buffer is only needed on the fp32-wire path
+        dq_buf = [torch.empty_like(dq)] if wire_dtype is None else []
         for r in range(1, W + 1):
             offset = get_partition_id(double_group, r)
             causal_shift = offset <= rank and r != 1  # q origin precedes
@@ -356,21 +412,29 @@ class OpBurstAttnStrip(torch.autograd.Function):
                 ring.double_ring_send_recv([dlt, grad_output, q, lse], read_bufs, r)
                 ring.commit()
             if r != 1:
-                dq_ring.double_ring_send_recv_q([dq], dq_buf, r)
+                if wire_dtype is not None:
+                    wire_s.copy_(dq)
+                    dq_ring.send_recv([wire_s], [wire_r])
+                else:
+                    dq_ring.double_ring_send_recv_q([dq], dq_buf, r)
                 dq_ring.commit()
-            delta = dlt if ctx.optimize_bwd_comm else P.bwd_preprocess(dlt, grad_output)
+            delta = (
+                dlt if ctx.optimize_bwd_comm
+                else P.bwd_preprocess(dlt, grad_output, out=delta_buf)
+            )
+            tgt = dq if r == 1 else dq_local
             if not causal_shift or not ctx.causal:
-                dq_i, dk_i, dv_i = P.bwd(
+                P.bwd_accum(
                     grad_output, q, k, v, delta, lse, scale, ctx.causal,
-                    ctx.deterministic,
+                    ctx.deterministic, tgt, dk, dv,
                 )
                 shifted = False
             else:
                 # shifted tile: q[1:] vs k[:-1] (reference :557-585)
-                dq_i, dk_i, dv_i = P.bwd(
+                P.bwd_accum(
                     grad_output[:, 1:], q[:, 1:], k[:, :-1], v[:, :-1],
                     delta[:, :, 1:], lse[:, :, 1:], scale, ctx.causal,
-                    ctx.deterministic,
+                    ctx.deterministic, tgt[:, 1:], dk[:, :-1], dv[:, :-1],
                 )
                 shifted = True
             if r != W:
@@ -382,20 +446,28 @@ class OpBurstAttnStrip(torch.autograd.Function):
             ring.wait()
             if r != 1:
                 dq_ring.wait()
-                recv, dq_buf = _record_stream(*dq_buf), [dq]
-                dq = recv[0]
-            if not shifted:
-                dq += dq_i
-                dk += dk_i
-                dv += dv_i
-            else:
-                dq[:, 1:] += dq_i
-                dk[:, :-1] += dk_i
-                dv[:, :-1] += dv_i
-        dq_ring.double_ring_send_recv_q([dq], dq_buf, W + 1)
-        dq_ring.commit()
-        dq_ring.wait()
-        dq = _record_stream(*dq_buf)[0]
+                if wire_dtype is not None:
+                    dq.copy_(wire_r)
+                else:
+                    recv, dq_buf = _record_stream(*dq_buf), [dq]
+                    dq = recv[0]
+                if shifted:
+                    dq[:, 1:] += dq_local[:, 1:]
+                    dq_local[:, 1:].zero_()
+                else:
+                    dq += dq_local
+                    dq_local.zero_()
+        if wire_dtype is not None:
+            wire_s.copy_(dq)
+            dq_ring.send_recv([wire_s], [wire_r])
+            dq_ring.commit()
+            dq_ring.wait()
+            dq.copy_(wire_r)
+        else:
+            dq_ring.double_ring_send_recv_q([dq], dq_buf, W + 1)
+            dq_ring.commit()
+            dq_ring.wait()
+            dq = _record_stream(*dq_buf)[0]
         return (
             dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype),
             None, None, None, None, None, None, None,

@@ -12,11 +12,20 @@ Provider contract (all tensors in flash layout [B, S, N, D]; lse/delta in
 
   fwd(q, k, v, scale, causal)            -> (o_i fp32 [B,Sq,N,D],
                                              lse_i fp32 [B,N,Sq])
-  bwd_preprocess(o, do)                  -> delta fp32 [B,N,S]
+  bwd_preprocess(o, do, out=None)        -> delta fp32 [B,N,S]
                                             (= rowsum(o*do); the flash bwd
                                             preprocess, cf. lao.py:247-269)
   bwd(do, q, k, v, delta, lse, scale,
       causal, deterministic)             -> (dq, dk, dv fp32 [.,.,N,D])
+  bwd_accum(do, q, k, v, delta, lse,
+      scale, causal, deterministic,
+      dq, dk, dv)                        -> None; ADDS the tile's dq/dk/dv
+                                            contribution into the given
+                                            fp32 accumulators (strided
+                                            views allowed) — the ring's
+                                            round accumulation runs in the
+                                            kernel epilogues, not python
+                                            adds
   merge(o, lse, o_i, lse_i)              -> merged (o, lse); o [B,S,N,D]
                                             fp32, lse [B,S,N,1] fp32,
                                             per burst_utils.py:20-33.
@@ -53,12 +62,19 @@ class HipTileProvider:
     def fwd(self, q, k, v, scale, causal):
         return self._ext.attn_fwd(q, k, v, float(scale), bool(causal))
 
-    def bwd_preprocess(self, o, do):
-        return self._ext.attn_bwd_preprocess(o, do)
+    def bwd_preprocess(self, o, do, out=None):
+        return self._ext.attn_bwd_preprocess(o, do, out)
 
     def bwd(self, do, q, k, v, delta, lse, scale, causal, deterministic):
         return self._ext.attn_bwd(
             do, q, k, v, delta, lse, float(scale), bool(causal), bool(deterministic)
+        )
+
+    def bwd_accum(self, do, q, k, v, delta, lse, scale, causal, deterministic,
+                  dq, dk, dv):
+        self._ext.attn_bwd_accum(
+            do, q, k, v, delta, lse, float(scale), bool(causal),
+            bool(deterministic), dq, dk, dv,
         )
 
     def merge(self, o, lse, o_i, lse_i):

@@ -21,7 +21,14 @@
  *     contiguous (flash-attn layout, burst_utils.py:150-163).
  *   - lse/delta inputs to the backward may be seq-sliced: strides are
  *     {batch, head} in elements, seq stride is 1.
- *   - dq/dk/dv outputs are fp32, contiguous [B, S*, N, D].
+ *   - dq/dk/dv outputs are fp32 ACCUMULATORS, strided {batch, seq, head}
+ *     (D contiguous): every call ADDS its tile contribution in place, so
+ *     the ring layer accumulates rounds without elementwise-add passes.
+ *     Pass zero-filled buffers for plain (non-accumulating) semantics.
+ *   - deterministic != 0 selects the atomic-free two-pass-dq kernel plan
+ *     (bitwise run-to-run identical); deterministic == 0 (default)
+ *     selects the fused dK+dQ plan with flash-attn's atomic fp32 dq
+ *     accumulation (2 fewer tile GEMMs).
  *   - causal implies Sq == Sk (equal-length tiles; the ring's zigzag /
  *     striped bookkeeping reduces every other case to non-causal tiles).
  *   - stream is a hipStream_t.
@@ -81,6 +88,8 @@ int bahip_attn_bwd(
     const int64_t do_strides[3], const int64_t q_strides[3],
     const int64_t k_strides[3], const int64_t v_strides[3],
     const int64_t delta_strides[2], const int64_t lse_strides[2],
+    const int64_t dq_strides[3], const int64_t dk_strides[3],
+    const int64_t dv_strides[3],
     float softmax_scale, int causal, int deterministic, int dtype,
     void* stream);
 

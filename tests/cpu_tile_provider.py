@@ -14,14 +14,26 @@ class OracleTileProvider:
     def fwd(self, q, k, v, scale, causal):
         return oracle.tile_fwd(q, k, v, scale, causal)
 
-    def bwd_preprocess(self, o, do):
+    def bwd_preprocess(self, o, do, out=None):
         # delta = rowsum(o * do) fp32, [B,S,N,D] -> [B,N,S]
-        return (o.to(torch.float32) * do.to(torch.float32)).sum(-1).transpose(1, 2).contiguous()
+        delta = (o.to(torch.float32) * do.to(torch.float32)).sum(-1).transpose(1, 2).contiguous()
+        if out is not None:
+            out.copy_(delta)
+            return out
+        return delta
 
     def bwd(self, do, q, k, v, delta, lse, scale, causal, deterministic):
         return oracle.tile_bwd(
             do, q, k, v, lse, scale, causal, softmax_d=delta
         )
+
+    def bwd_accum(self, do, q, k, v, delta, lse, scale, causal, deterministic,
+                  dq, dk, dv):
+        dq_i, dk_i, dv_i = self.bwd(do, q, k, v, delta, lse, scale, causal,
+                                    deterministic)
+        dq += dq_i
+        dk += dk_i
+        dv += dv_i
 
     def merge(self, o, lse, o_i, lse_i):
         return oracle.scale_out_lse(o, lse, o_i, lse_i)

@@ -291,8 +291,8 @@ def test_interface_flags_single_rank(opt_bwd, det):
 
 
 def test_backward_bitwise_deterministic():
-    """The two-pass backward has no atomics — repeated runs must be
-    BITWISE identical (both deterministic=True and False)."""
+    """deterministic=True selects the atomic-free two-pass plan: repeated
+    runs must be BITWISE identical."""
     b, s, n, d = 1, 1024, 4, 128
     dtype = torch.float16
     q = _rand(b, s, n, d, dtype, 91)
@@ -303,12 +303,66 @@ def test_backward_bitwise_deterministic():
     scale = 1.0 / math.sqrt(d)
     o, lse = ext.attn_fwd(q, k, v, scale, True)
     delta = ext.attn_bwd_preprocess(o.to(dtype), do)
-    outs = []
-    for det in (False, True, False):
-        outs.append(ext.attn_bwd(do, q, k, v, delta, lse, scale, True, det))
+    outs = [
+        ext.attn_bwd(do, q, k, v, delta, lse, scale, True, True)
+        for _ in range(3)
+    ]
     for i in (1, 2):
         for a, b_ in zip(outs[0], outs[i]):
             assert torch.equal(a, b_), "backward is not bitwise deterministic"
+
+
+@pytest.mark.parametrize("causal", [False, True])
+@pytest.mark.parametrize("dtype", [torch.float16, torch.bfloat16])
+def test_bwd_fused_vs_split(causal, dtype):
+    """The fused dK+dQ plan (deterministic=False) must agree with the
+    atomic-free split plan (deterministic=True) — both accumulate fp32;
+    the only differences are summation order and the dS scratch round
+    trip, so the gap must be far below the oracle tolerance."""
+    b, s, n, d = 1, 768, 3, 128
+    q = _rand(b, s, n, d, dtype, 101)
+    k = _rand(b, s, n, d, dtype, 102)
+    v = _rand(b, s, n, d, dtype, 103)
+    do = _rand(b, s, n, d, dtype, 104)
+    ext = _ext()
+    scale = 1.0 / math.sqrt(d)
+    o, lse = ext.attn_fwd(q, k, v, scale, causal)
+    delta = ext.attn_bwd_preprocess(o.to(dtype), do)
+    fused = ext.attn_bwd(do, q, k, v, delta, lse, scale, causal, False)
+    split = ext.attn_bwd(do, q, k, v, delta, lse, scale, causal, True)
+    for f, s_, name in zip(fused, split, ("dq", "dk", "dv")):
+        torch.testing.assert_close(f, s_, rtol=2e-3, atol=1e-3, msg=name)
+
+
+@pytest.mark.parametrize("det", [False, True])
+def test_bwd_accum_strided_views(det):
+    """bwd_accum ADDS into strided accumulator views (the zigzag
+    half-slice targets): accumulating the same tile twice into a
+    half-slice view must double exactly that region and leave the rest
+    untouched."""
+    b, s, n, d = 1, 512, 2, 128
+    dtype = torch.float16
+    half = s // 2
+    q = _rand(b, s, n, d, dtype, 111)
+    k = _rand(b, s, n, d, dtype, 112)
+    v = _rand(b, s, n, d, dtype, 113)
+    do = _rand(b, s, n, d, dtype, 114)
+    ext = _ext()
+    scale = 1.0 / math.sqrt(d)
+    o, lse = ext.attn_fwd(q[:, half:], k, v, scale, False)
+    delta = ext.attn_bwd_preprocess(o.to(dtype), do[:, half:])
+    ref = ext.attn_bwd(do[:, half:], q[:, half:], k, v, delta, lse, scale,
+                       False, det)
+    dq = torch.zeros(b, s, n, d, dtype=torch.float32, device="cuda")
+    dk = torch.zeros_like(dq)
+    dv = torch.zeros_like(dq)
+    for _ in range(2):
+        ext.attn_bwd_accum(do[:, half:], q[:, half:], k, v, delta, lse,
+                           scale, False, det, dq[:, half:], dk, dv)
+    assert torch.all(dq[:, :half] == 0), "untouched region modified"
+    torch.testing.assert_close(dq[:, half:], 2 * ref[0], rtol=1e-4, atol=1e-4)
+    torch.testing.assert_close(dk, 2 * ref[1], rtol=1e-4, atol=1e-4)
+    torch.testing.assert_close(dv, 2 * ref[2], rtol=1e-4, atol=1e-4)
 
 
 def test_full_size_properties():
