@@ -536,12 +536,39 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
 //     register-resident exactly like the split kernel.
 // FUSE_DQ=0 degenerates to a flipped dK-only kernel (atomic-free — legal
 // for deterministic=1; kept for A/B against the split MODE 1).
-template <typename T, int D, int FUSE_DQ>
+//
+// Atomic-contention control: the grid is launched HEAD-FIRST
+// (blockIdx.x = head) so concurrent workgroups flush different heads'
+// dq regions, and non-causal workgroups start their q-tile stream at a
+// per-kv-block offset (wrapping) so same-head workgroups flush
+// DIFFERENT q tiles at any instant — without both, the 256 same-head
+// workgroups march the same 16 KB of dq lines in lockstep and the
+// line-serialised L2 RMWs dominate (measured 18x).
+// DBG (env BA_DKQ_DBG, perf probes only — results wrong for >0):
+//   1 = skip the global atomic flush;  2 = plain LDS writes, no ds_add;
+//   3 = skip the whole dQ section (KT staging/LDS size/barriers kept);
+//   4 = dQ MFMAs with dsT as both operands (no K^T LDS reads);
+//   5 = dQ MFMAs + K^T reads, but no ds_add and no flush.
+// SCRW: dS scratch row width (64 = conflict-free reads, 32 = half LDS).
+//
+// FUSE_DV=1 (requires FUSE_DQ=0): the 4-GEMM fused dK+dV kernel — adds
+// P (kept beside dS), a streamed dO^T image, and a second accumulator:
+//   dV^T? no — dV[kv][d] += mfma(P-frags [row=kv][contr=q],
+//                                dO^T image [row=d][contr=q])
+// P and dS share ONE per-wave scratch region sequentially (DS ops retire
+// in order within a wave, so the write-read-write-read chain is safe
+// without barriers).  This replaces BOTH split dkv kernels with 4 tile
+// GEMMs instead of 5, atomic-free and deterministic — the measured
+// verdict on gfx950 is that fp32 atomics (LDS ds_add ~2.6 s, global L2
+// ~1.8 s at this density, each atomic dropping its L2 line) rule out
+// flash-attn's atomic dq plan entirely (tools/dkq_probe.hip).
+template <typename T, int D, int FUSE_DQ, int DBG = 0, int SCRW = 64,
+          int FUSE_DV = 0>
 __global__ __launch_bounds__(512) void bwd_dkq_kernel(
     const T* __restrict__ dout, const T* __restrict__ q,
     const T* __restrict__ k, const T* __restrict__ v,
     const float* __restrict__ delta, const float* __restrict__ lse,
-    float* __restrict__ dq, float* __restrict__ dk,
+    float* __restrict__ dq, float* __restrict__ dk, float* __restrict__ dv,
     int Sq, int Sk, int N,
     int64_t g_sb, int64_t g_ss, int64_t g_sh,
     int64_t q_sb, int64_t q_ss, int64_t q_sh,
@@ -550,43 +577,63 @@ __global__ __launch_bounds__(512) void bwd_dkq_kernel(
     int64_t d_sb, int64_t d_sh, int64_t l_sb, int64_t l_sh,
     int64_t dq_sb, int64_t dq_ss, int64_t dq_sh,
     int64_t dk_sb, int64_t dk_ss, int64_t dk_sh,
+    int64_t dv_sb, int64_t dv_ss, int64_t dv_sh,
     float scale, int causal) {
+  static_assert(!(FUSE_DQ && FUSE_DV), "dq and dv fusions are exclusive");
   using MT = mfma_traits<T>;
   using frag = typename MT::frag;
   constexpr int QBLK = 32;    // streamed q rows per tile (1 subtile)
   constexpr int KVWG = 256;   // kv columns per workgroup (8 waves x 32)
   constexpr int NT = 512;
+  constexpr int NIMG = 3 + FUSE_DV;         // Q, dO, Q^T [, dO^T]
   constexpr int SWZ = (D == 128) ? 15 : 7;  // Q/dO row-major images
   constexpr int SWZ_QT = 3;                 // Q^T image rows are 64 B
-  constexpr int SCRW = 64;  // scratch row width (elems; 32 used) -> 128 B
-  constexpr int SWZ_SC = 7;                 // conflict-free dsf reads
+  constexpr int SWZ_SC = (SCRW == 64) ? 7 : 3;
   constexpr int CHUNKS = QBLK * (D / 8);    // staged 8-elem chunks per img
 
   // one LDS object (a second __shared__ forces vmcnt(0) per ds_read):
-  // [2 buf x {Q row-major | dO row-major | Q^T}] [K^T persistent]
-  // [8 x per-wave dS scratch] [fp32 dq reduce buffer]
-  constexpr int STREAM = 2 * 3 * QBLK * D;
+  // [2 buf x {Q row-major | dO row-major | Q^T | (dO^T)}] [K^T persistent]
+  // [V row-major persistent (FUSE_DV — keeps the V fragments out of the
+  // register file; kv-resident so staged once)] [8 x per-wave P/dS
+  // scratch] [fp32 dq reduce buffer]
+  constexpr int STREAM = 2 * NIMG * QBLK * D;
   constexpr int KT_E = FUSE_DQ ? D * KVWG : 0;
+  constexpr int V_E = FUSE_DV ? D * KVWG : 0;
   constexpr int SCR_E = 8 * 32 * SCRW;
+  constexpr int SCR2_E = FUSE_DV ? SCR_E : 0;  // separate P scratch
   constexpr int REDU_E = FUSE_DQ ? QBLK * D * (int)(4 / sizeof(T)) : 0;
-  __shared__ T lds[STREAM + KT_E + SCR_E + REDU_E];
-  auto ldsQ = [&](int buf) -> T* { return lds + buf * (3 * QBLK * D); };
+  static_assert((STREAM + KT_E + V_E + SCR_E + SCR2_E + REDU_E) *
+                        (int)sizeof(T) <=
+                    163840,
+                "LDS budget exceeded");
+  __shared__ T lds[STREAM + KT_E + V_E + SCR_E + SCR2_E + REDU_E];
+  auto ldsQ = [&](int buf) -> T* { return lds + buf * (NIMG * QBLK * D); };
   auto ldsG = [&](int buf) -> T* {
-    return lds + buf * (3 * QBLK * D) + QBLK * D;
+    return lds + buf * (NIMG * QBLK * D) + QBLK * D;
   };
   auto ldsQT = [&](int buf) -> T* {
-    return lds + buf * (3 * QBLK * D) + 2 * QBLK * D;
+    return lds + buf * (NIMG * QBLK * D) + 2 * QBLK * D;
+  };
+  auto ldsGT = [&](int buf) -> T* {  // FUSE_DV only
+    return lds + buf * (NIMG * QBLK * D) + 3 * QBLK * D;
   };
   T* ldsKT = lds + STREAM;
-  auto ldsSC = [&](int w) -> T* { return lds + STREAM + KT_E + w * 32 * SCRW; };
-  float* redu = (float*)(lds + STREAM + KT_E + SCR_E);
+  T* ldsV = lds + STREAM + KT_E;
+  auto ldsSC = [&](int w) -> T* {
+    return lds + STREAM + KT_E + V_E + w * 32 * SCRW;
+  };
+  auto ldsSC2 = [&](int w) -> T* {  // FUSE_DV: P scratch
+    return lds + STREAM + KT_E + V_E + SCR_E + w * 32 * SCRW;
+  };
+  float* redu = (float*)(lds + STREAM + KT_E + V_E + SCR_E + SCR2_E);
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
   const int l31 = lane & 31, hi = lane >> 5;
-  const int n = blockIdx.y, b = blockIdx.z;
-  const int kvb = blockIdx.x * KVWG + wave * 32;  // wave's kv block
+  const int n = blockIdx.x, b = blockIdx.z;     // head-first dispatch
+  const int kvblk = blockIdx.y;
+  const int kvb = kvblk * KVWG + wave * 32;     // wave's kv block
   const int kv_col = kvb + l31;
 
   const T* qp = q + b * q_sb + (int64_t)n * q_sh;
@@ -596,19 +643,23 @@ __global__ __launch_bounds__(512) void bwd_dkq_kernel(
   const float* dp_ = delta + b * d_sb + n * d_sh;
   const float* lp_ = lse + b * l_sb + n * l_sh;
 
-  // resident K and V fragments [row=kv][contr=d] (A operands of S^T/dP^T)
-  frag kf[D / 16], vf[D / 16];
+  // resident K fragments [row=kv][contr=d] (A operand of S^T); V the
+  // same way in registers UNLESS FUSE_DV (register budget: it moves to
+  // the persistent LDS image instead)
+  frag kf[D / 16], vf[FUSE_DV ? 1 : D / 16];
 #pragma unroll
   for (int s = 0; s < D / 16; ++s) {
     if (kv_col < Sk) {
       kf[s] = __builtin_bit_cast(
           frag, *(const u32x4_t*)(kp + (int64_t)kv_col * k_ss + 16 * s + 8 * hi));
-      vf[s] = __builtin_bit_cast(
-          frag, *(const u32x4_t*)(vp + (int64_t)kv_col * v_ss + 16 * s + 8 * hi));
+      if (!FUSE_DV)
+        vf[s] = __builtin_bit_cast(
+            frag,
+            *(const u32x4_t*)(vp + (int64_t)kv_col * v_ss + 16 * s + 8 * hi));
     } else {
       u32x4_t z = {0, 0, 0, 0};
       kf[s] = __builtin_bit_cast(frag, z);
-      vf[s] = __builtin_bit_cast(frag, z);
+      if (!FUSE_DV) vf[s] = __builtin_bit_cast(frag, z);
     }
   }
   const float c2 = scale * BA_LOG2E;
@@ -616,6 +667,10 @@ __global__ __launch_bounds__(512) void bwd_dkq_kernel(
   f32x16_t acc[D / 32];  // dK^T accumulator [d rows][kv col = lane]
 #pragma unroll
   for (int dt = 0; dt < D / 32; ++dt) acc[dt] = (f32x16_t)(0.f);
+  f32x16_t acc2[FUSE_DV ? D / 32 : 1];  // dV accumulator [kv rows][d col]
+#pragma unroll
+  for (int dt = 0; dt < (FUSE_DV ? D / 32 : 1); ++dt)
+    acc2[dt] = (f32x16_t)(0.f);
 
   // one-time: zero the reduce buffer; stage K^T [D rows][KVWG] transposed
   if (FUSE_DQ) {
@@ -625,15 +680,36 @@ __global__ __launch_bounds__(512) void bwd_dkq_kernel(
     for (int c = 0; c < PTK; ++c) {
       const int flat = tid + c * NT;
       const int kv = flat / (D / 8), col8 = flat % (D / 8);
-      const int kvg = blockIdx.x * KVWG + kv;
+      const int kvg = kvblk * KVWG + kv;
       const int kvc = kvg < Sk ? kvg : (Sk - 1);
       u32x4_t ch = *(const u32x4_t*)(kp + (int64_t)kvc * k_ss + col8 * 8);
       ba_st_transposed<T, KVWG, 15, 0>(ldsKT, kv, col8 * 8, ch);
     }
   }
+  // one-time (FUSE_DV): stage the workgroup's V block row-major
+  if (FUSE_DV) {
+    constexpr int PTV = (KVWG * D / 8) / NT;
+#pragma unroll
+    for (int c = 0; c < PTV; ++c) {
+      const int flat = tid + c * NT;
+      const int kv = flat / (D / 8), col8 = flat % (D / 8);
+      const int kvg = kvblk * KVWG + kv;
+      const int kvc = kvg < Sk ? kvg : (Sk - 1);
+      u32x4_t ch = *(const u32x4_t*)(vp + (int64_t)kvc * v_ss + col8 * 8);
+      const int byte = ba_swz<SWZ>(kv * (2 * D) + col8 * 16, kv);
+      *(u32x4_t*)((char*)ldsV + byte) = ch;
+    }
+  }
 
-  const int t0 = causal ? (blockIdx.x * KVWG) / QBLK : 0;
+  const int t0 = causal ? (kvblk * KVWG) / QBLK : 0;
   const int nt = (Sq + QBLK - 1) / QBLK;
+  const int span = nt - t0;
+  // non-causal stagger: spread same-head workgroups evenly over the
+  // q-tile sequence (causal blocks are naturally staggered by t0)
+  const int stag =
+      (FUSE_DQ && !causal && gridDim.y > 1)
+          ? (int)(((int64_t)kvblk * span) / gridDim.y)
+          : 0;
 
   // per-tile prefetch: Q and dO chunks + this lane's lse*log2e and delta
   auto issue_loads = [&](int tile, u32x4_t* qreg, u32x4_t* greg, float* lse2,
@@ -668,6 +744,9 @@ __global__ __launch_bounds__(512) void bwd_dkq_kernel(
         *(u32x4_t*)((char*)ldsG(buf) + byte) = greg[c];
         ba_st_transposed<T, QBLK, SWZ_QT, 3>(ldsQT(buf), row, col8 * 8,
                                              qreg[c]);
+        if (FUSE_DV)
+          ba_st_transposed<T, QBLK, SWZ_QT, 3>(ldsGT(buf), row, col8 * 8,
+                                               greg[c]);
       }
     }
   };
@@ -675,7 +754,7 @@ __global__ __launch_bounds__(512) void bwd_dkq_kernel(
   float lse2_c, dlt_c;
   {
     u32x4_t qreg[(CHUNKS + NT - 1) / NT], greg[(CHUNKS + NT - 1) / NT];
-    issue_loads(t0, qreg, greg, &lse2_c, &dlt_c);
+    issue_loads(t0 + stag, qreg, greg, &lse2_c, &dlt_c);
     write_lds(0, qreg, greg);
     __syncthreads();
   }
@@ -683,12 +762,15 @@ __global__ __launch_bounds__(512) void bwd_dkq_kernel(
     __builtin_amdgcn_s_setprio(1);  // T5 static form (younger half)
 
   int cur = 0;
-  for (int t = t0; t < nt; ++t) {
+  int t = t0 + stag;  // wrapped iteration order over [t0, nt)
+  for (int i = 0; i < span; ++i) {
     const int q0 = t * QBLK;
-    const bool has_next = (t + 1) < nt;
+    int tn = t + 1;
+    if (tn >= nt) tn = t0;
+    const bool has_next = (i + 1) < span;
     u32x4_t qreg[(CHUNKS + NT - 1) / NT], greg[(CHUNKS + NT - 1) / NT];
     float lse2_n = 0.f, dlt_n = 0.f;
-    if (has_next) issue_loads(t + 1, qreg, greg, &lse2_n, &dlt_n);
+    if (has_next) issue_loads(tn, qreg, greg, &lse2_n, &dlt_n);
 
     // waves whose kv block is wholly after this q tile are masked anyway
     const bool active = !causal || (q0 + QBLK - 1 >= kvb);
@@ -699,10 +781,26 @@ __global__ __launch_bounds__(512) void bwd_dkq_kernel(
       for (int s = 0; s < D / 16; ++s) {
         frag qfr = ba_ld_rowslice<T, D, SWZ>(ldsQ(cur), l31, 16 * s + 8 * hi);
         frag gfr = ba_ld_rowslice<T, D, SWZ>(ldsG(cur), l31, 16 * s + 8 * hi);
+        frag vop;
+        if (FUSE_DV)
+          vop = ba_ld_rowslice<T, D, SWZ>(ldsV, wave * 32 + l31,
+                                          16 * s + 8 * hi);
+        else
+          vop = vf[s];
         st = MT::mma(kf[s], qfr, st);
-        dpt = MT::mma(vf[s], gfr, dpt);
+        dpt = MT::mma(vop, gfr, dpt);
       }
-      // ---- dS^T = p * (dP^T - delta) * scale, lse/delta LANE-LOCAL
+      // ---- dS^T = p * (dP^T - delta) * scale, lse/delta LANE-LOCAL.
+      // FUSE_DV streams BOTH P and dS straight into their own per-wave
+      // scratch regions inside this loop (no register homes, no
+      // write-read-write serialisation): the dV/dK fragments read them
+      // back transposed.
+      T* sc = ldsSC(wave);
+      T* sc2 = ldsSC2(wave);
+      auto scr_byte = [&](int r) {
+        const int row = ba_crow(r, 0) + 4 * hi;  // kv-local
+        return ba_swz<SWZ_SC, 0>(row * (2 * SCRW) + 2 * l31, row);
+      };
       const int q_g = q0 + l31;
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
@@ -711,19 +809,19 @@ __global__ __launch_bounds__(512) void bwd_dkq_kernel(
             q_g < Sq && kv_g < Sk && (!causal || q_g >= kv_g);
         const float e = valid ? __builtin_fmaf(st[r], c2, -lse2_c) : BA_NEG_BIG;
         const float p = ba_exp2(e);
+        if (FUSE_DV) *(T*)((char*)sc2 + scr_byte(r)) = (T)p;
         st[r] = p * (dpt[r] - dlt_c) * scale;
       }
-      // ---- per-wave scratch transpose: dS^T D-layout -> dS fragments
-      {
-        T* sc = ldsSC(wave);
+      auto scr_put = [&](const f32x16_t& vv) {
 #pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          const int row = ba_crow(r, 0) + 4 * hi;  // kv-local
-          const int byte = ba_swz<SWZ_SC, 0>(row * (2 * SCRW) + 2 * l31, row);
-          *(T*)((char*)sc + byte) = (T)st[r];
-        }
-      }
-      if (FUSE_DQ) {
+        for (int r = 0; r < 16; ++r)
+          *(T*)((char*)sc + scr_byte(r)) = (T)vv[r];
+      };
+      // ---- per-wave scratch transpose: dS^T D-layout -> dS fragments
+      // (own region — the dV section's P reads don't serialise against
+      // these writes)
+      scr_put(st);
+      if (FUSE_DQ && DBG != 3) {
         // ---- dQ: fragments [row=q][contr=kv] in-register; one dt tile
         // at a time to keep the partial's register span at 16
         frag dsT[2];
@@ -733,24 +831,45 @@ __global__ __launch_bounds__(512) void bwd_dkq_kernel(
           f32x16_t dqp = (f32x16_t)(0.f);
 #pragma unroll
           for (int u = 0; u < 2; ++u) {
-            frag ktf = ba_ld_rowslice<T, KVWG, 15, 0>(
-                ldsKT, dt * 32 + l31, wave * 32 + 16 * u + 8 * hi);
+            frag ktf;
+            if (DBG == 4)  // perf probe: no K^T LDS reads
+              ktf = dsT[u];
+            else
+              ktf = ba_ld_rowslice<T, KVWG, 15, 0>(
+                  ldsKT, dt * 32 + l31, wave * 32 + 16 * u + 8 * hi);
             dqp = MT::mma(dsT[u], ktf, dqp);
           }
 #pragma unroll
           for (int r = 0; r < 16; ++r) {
             const int q_loc = ba_crow(r, 0) + 4 * hi;
-            atomicAdd(&redu[q_loc * D + dt * 32 + l31], dqp[r]);  // ds_add
+            if (DBG == 5) {         // perf probe: no LDS reduce at all
+              if (dqp[r] == 1234.5678f) redu[0] = dqp[r];  // keep dqp live
+            } else if (DBG == 2)    // perf probe: write, not LDS-atomic
+              redu[q_loc * D + dt * 32 + l31] = dqp[r];
+            else
+              atomicAdd(&redu[q_loc * D + dt * 32 + l31], dqp[r]);  // ds_add
+          }
+        }
+      }
+      if (FUSE_DV) {
+        // ---- dV += mfma(P, dO^T): P read back from its own scratch
+        frag pf[2];
+        pf[0] = ba_ld_rowslice<T, SCRW, SWZ_SC, 0>(sc2, l31, 8 * hi);
+        pf[1] = ba_ld_rowslice<T, SCRW, SWZ_SC, 0>(sc2, l31, 16 + 8 * hi);
+#pragma unroll
+        for (int dt = 0; dt < D / 32; ++dt) {
+#pragma unroll
+          for (int u = 0; u < 2; ++u) {
+            frag gt = ba_ld_rowslice<T, QBLK, SWZ_QT, 3>(
+                ldsGT(cur), dt * 32 + l31, 16 * u + 8 * hi);
+            acc2[dt] = MT::mma(pf[u], gt, acc2[dt]);
           }
         }
       }
       // ---- dK^T += mfma(Q^T, dS)
       frag dsf[2];
-      {
-        const T* sc = ldsSC(wave);
-        dsf[0] = ba_ld_rowslice<T, SCRW, SWZ_SC, 0>(sc, l31, 8 * hi);
-        dsf[1] = ba_ld_rowslice<T, SCRW, SWZ_SC, 0>(sc, l31, 16 + 8 * hi);
-      }
+      dsf[0] = ba_ld_rowslice<T, SCRW, SWZ_SC, 0>(sc, l31, 8 * hi);
+      dsf[1] = ba_ld_rowslice<T, SCRW, SWZ_SC, 0>(sc, l31, 16 + 8 * hi);
 #pragma unroll
       for (int dt = 0; dt < D / 32; ++dt) {
 #pragma unroll
@@ -761,15 +880,19 @@ __global__ __launch_bounds__(512) void bwd_dkq_kernel(
         }
       }
     }
-    if (has_next) write_lds(cur ^ 1, qreg, greg);
-    __syncthreads();
-    if (FUSE_DQ) {
+    if (!FUSE_DQ) {
+      if (has_next) write_lds(cur ^ 1, qreg, greg);
+      __syncthreads();
+    } else {
+      __syncthreads();  // A: every wave's ds_adds for this tile landed
       // flush the reduced dq partial: ONE atomic per element per
-      // workgroup (the ds_adds above summed the 8 waves' 32-kv partials)
+      // workgroup (the ds_adds above summed the 8 waves' 32-kv
+      // partials).  Issued BEFORE write_lds so the staging's counted
+      // vmcnt wait does not drain the fire-and-forget atomics.
       const int base = tid * (QBLK * D / NT);  // 8 consecutive floats
       const int q_loc = base / D, d0 = base % D;
       const int q_g2 = q0 + q_loc;
-      if (q_g2 < Sq) {
+      if (q_g2 < Sq && DBG != 1 && DBG != 5) {
         float* gdst = dq + b * dq_sb + (int64_t)q_g2 * dq_ss + n * dq_sh + d0;
 #pragma unroll
         for (int j = 0; j < QBLK * D / NT; ++j) {
@@ -780,11 +903,13 @@ __global__ __launch_bounds__(512) void bwd_dkq_kernel(
 #pragma unroll
         for (int j = 0; j < QBLK * D / NT; ++j) redu[base + j] = 0.f;
       }
-      __syncthreads();
+      if (has_next) write_lds(cur ^ 1, qreg, greg);
+      __syncthreads();  // B: redu zeroed + staging visible
     }
     cur ^= 1;
     lse2_c = lse2_n;
     dlt_c = dlt_n;
+    t = tn;
   }
 
   if (kv_col < Sk) {
@@ -794,6 +919,20 @@ __global__ __launch_bounds__(512) void bwd_dkq_kernel(
 #pragma unroll
       for (int r = 0; r < 16; ++r)
         row[dt * 32 + ba_crow(r, hi)] += acc[dt][r];
+  }
+  if (FUSE_DV) {
+    // dV accumulator sits [kv rows][d col = lane]: per (r, dt) the 32
+    // lanes of a half write one kv row's 128 B contiguously
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int kv_g = kvb + ba_crow(r, 0) + 4 * hi;
+      if (kv_g < Sk) {
+        float* row2 = dv + b * dv_sb + (int64_t)kv_g * dv_ss + n * dv_sh;
+#pragma unroll
+        for (int dt = 0; dt < D / 32; ++dt)
+          row2[dt * 32 + l31] += acc2[dt][r];
+      }
+    }
   }
 }
 
@@ -821,12 +960,24 @@ extern "C" int bahip_attn_bwd_preprocess(
   return 0;
 }
 
-// plan selection: deterministic=0 -> fused dK+dQ (6 tile GEMMs, atomic
-// dq) + dV; deterministic=1 -> atomic-free split 3-kernel plan.
-// BA_BWD_FUSED=0/1 forces the plan (A/B knob; 1 with deterministic=1 is
-// refused — the fused dq accumulation order is not fixed).
-// BA_BWD_DK_FLIP=1 swaps the split plan's MODE-1 dK kernel for the
-// flipped (FUSE_DQ=0) kernel — atomic-free, so deterministic-legal.
+// plan selection (all plans accumulate into dq/dk/dv; plans 0 and 1 are
+// bitwise deterministic — the `deterministic` flag is honoured by
+// construction):
+//   plan 0 "split8" (DEFAULT): dq kernel + split dV (MODE 0) + split dK
+//     (MODE 1) — 8 tile GEMMs, two-pass dq, atomic-free.  Fastest
+//     measured plan on gfx950 (tools/dkq_probe.hip round 2).
+//     BA_BWD_DK_FLIP=1 swaps MODE 1 for the flipped FUSE_DQ=0 kernel
+//     (measured equal, 142 vs 137 ms at s=65536).
+//   plan 1 "dkvf7" (BA_BWD_FUSED=1): dq kernel + fused flipped dK+dV —
+//     7 tile GEMMs, atomic-free, but the QBLK=32 single-workgroup
+//     structure + P/dS scratch round trips measured SLOWER than split8
+//     (260 vs 230 ms for the dK+dV part at s=65536); kept as the
+//     documented experiment.
+//   plan 2 "atomic6" (BA_BWD_FUSED=2): flash-attn's plan — fused dK+dQ
+//     with fp32 atomic dq + split dV.  6 GEMMs but MEASURED DEAD on
+//     gfx950 (tools/dkq_probe.hip: LDS ds_add ~2.6 s, global fp32
+//     atomics ~1.8 s at this density — atomics drop their L2 line);
+//     kept for the record and for re-testing on future silicon.
 template <typename T, int D>
 static int launch_bwd(const void* dout, const void* q, const void* k,
                       const void* v, const float* delta, const float* lse,
@@ -837,19 +988,28 @@ static int launch_bwd(const void* dout, const void* q, const void* k,
                       const int64_t* dqs, const int64_t* dks,
                       const int64_t* dvs, float scale, int causal,
                       int deterministic, void* stream) {
-  static const int env_fused = [] {
-    const char* e = getenv("BA_BWD_FUSED");
-    return e ? atoi(e) : -1;
-  }();
-  static const int dk_flip = [] {
-    const char* e = getenv("BA_BWD_DK_FLIP");
-    return e ? atoi(e) : 0;
-  }();
-  const bool fused = env_fused >= 0 ? (env_fused && !deterministic)
-                                    : !deterministic;
+  // read per call (cheap at one bwd invocation) so tests can flip plans
+  const char* ef = getenv("BA_BWD_FUSED");
+  const int env_fused = ef ? atoi(ef) : -1;
+  const char* df = getenv("BA_BWD_DK_FLIP");
+  const int dk_flip = df ? atoi(df) : 0;
+  const char* dbg = getenv("BA_DKQ_DBG");
+  const int dkq_dbg = dbg ? atoi(dbg) : 0;
+  (void)deterministic;  // plans 0/1 are deterministic by construction
+  const int plan = env_fused >= 0 ? env_fused : 0;
   dim3 grid_kv((unsigned)((Sk + 255) / 256), (unsigned)N, (unsigned)B);
-  if (!fused) {
-    dim3 grid_dq((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B);
+  // head-first grid for the dkq/dkvf kernels
+  dim3 grid_dkq((unsigned)N, (unsigned)((Sk + 255) / 256), (unsigned)B);
+  dim3 grid_dq((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B);
+
+#define DKQ_ARGS                                                              \
+  (const T*)dout, (const T*)q, (const T*)k, (const T*)v, delta, lse, dq, dk, \
+      dv, (int)Sq, (int)Sk, (int)N, gs[0], gs[1], gs[2], qs[0], qs[1],        \
+      qs[2], ks[0], ks[1], ks[2], vs[0], vs[1], vs[2], ds[0], ds[1], ls[0],   \
+      ls[1], dqs[0], dqs[1], dqs[2], dks[0], dks[1], dks[2], dvs[0], dvs[1],  \
+      dvs[2], scale, causal
+
+  if (plan != 2) {
     bwd_dq_kernel<T, D><<<grid_dq, 512, 0, (hipStream_t)stream>>>(
         (const T*)dout, (const T*)q, (const T*)k, (const T*)v, delta, lse, dq,
         (int)Sq, (int)Sk, (int)N, gs[0], gs[1], gs[2], qs[0], qs[1], qs[2],
@@ -857,25 +1017,31 @@ static int launch_bwd(const void* dout, const void* q, const void* k,
         dqs[0], dqs[1], dqs[2], scale, causal);
     BA_CHECK_LAUNCH();
   }
+  if (plan == 1) {
+    bwd_dkq_kernel<T, D, 0, 0, 32, 1>
+        <<<grid_dkq, 512, 0, (hipStream_t)stream>>>(DKQ_ARGS);
+    BA_CHECK_LAUNCH();
+    return 0;
+  }
   bwd_dkv_kernel<T, D, 0, 64><<<grid_kv, 512, 0, (hipStream_t)stream>>>(
       (const T*)dout, (const T*)q, (const T*)k, (const T*)v, delta, lse, dv,
       nullptr, (int)Sq, (int)Sk, (int)N, gs[0], gs[1], gs[2], qs[0], qs[1],
       qs[2], ks[0], ks[1], ks[2], vs[0], vs[1], vs[2], ds[0], ds[1], ls[0],
       ls[1], dvs[0], dvs[1], dvs[2], scale, causal);
   BA_CHECK_LAUNCH();
-  if (fused) {
-    bwd_dkq_kernel<T, D, 1><<<grid_kv, 512, 0, (hipStream_t)stream>>>(
-        (const T*)dout, (const T*)q, (const T*)k, (const T*)v, delta, lse,
-        dq, dk, (int)Sq, (int)Sk, (int)N, gs[0], gs[1], gs[2], qs[0], qs[1],
-        qs[2], ks[0], ks[1], ks[2], vs[0], vs[1], vs[2], ds[0], ds[1], ls[0],
-        ls[1], dqs[0], dqs[1], dqs[2], dks[0], dks[1], dks[2], scale, causal);
+  if (plan == 2) {
+    if (dkq_dbg == 1)
+      bwd_dkq_kernel<T, D, 1, 1>
+          <<<grid_dkq, 512, 0, (hipStream_t)stream>>>(DKQ_ARGS);
+    else if (dkq_dbg == 2)
+      bwd_dkq_kernel<T, D, 1, 2>
+          <<<grid_dkq, 512, 0, (hipStream_t)stream>>>(DKQ_ARGS);
+    else
+      bwd_dkq_kernel<T, D, 1, 0>
+          <<<grid_dkq, 512, 0, (hipStream_t)stream>>>(DKQ_ARGS);
     BA_CHECK_LAUNCH();
   } else if (dk_flip) {
-    bwd_dkq_kernel<T, D, 0><<<grid_kv, 512, 0, (hipStream_t)stream>>>(
-        (const T*)dout, (const T*)q, (const T*)k, (const T*)v, delta, lse,
-        dq, dk, (int)Sq, (int)Sk, (int)N, gs[0], gs[1], gs[2], qs[0], qs[1],
-        qs[2], ks[0], ks[1], ks[2], vs[0], vs[1], vs[2], ds[0], ds[1], ls[0],
-        ls[1], dqs[0], dqs[1], dqs[2], dks[0], dks[1], dks[2], scale, causal);
+    bwd_dkq_kernel<T, D, 0><<<grid_dkq, 512, 0, (hipStream_t)stream>>>(DKQ_ARGS);
     BA_CHECK_LAUNCH();
   } else {
     bwd_dkv_kernel<T, D, 1><<<grid_kv, 512, 0, (hipStream_t)stream>>>(
@@ -885,6 +1051,7 @@ static int launch_bwd(const void* dout, const void* q, const void* k,
         ls[0], ls[1], dks[0], dks[1], dks[2], scale, causal);
     BA_CHECK_LAUNCH();
   }
+#undef DKQ_ARGS
   return 0;
 }
 

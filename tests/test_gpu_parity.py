@@ -312,13 +312,16 @@ def test_backward_bitwise_deterministic():
             assert torch.equal(a, b_), "backward is not bitwise deterministic"
 
 
+@pytest.mark.parametrize("plan", ["1", "2"])
 @pytest.mark.parametrize("causal", [False, True])
 @pytest.mark.parametrize("dtype", [torch.float16, torch.bfloat16])
-def test_bwd_fused_vs_split(causal, dtype):
-    """The fused dK+dQ plan (deterministic=False) must agree with the
-    atomic-free split plan (deterministic=True) — both accumulate fp32;
-    the only differences are summation order and the dS scratch round
-    trip, so the gap must be far below the oracle tolerance."""
+def test_bwd_alt_plans_vs_split(plan, causal, dtype, monkeypatch):
+    """The experimental kernel plans (BA_BWD_FUSED=1 fused dK+dV, =2
+    flash-attn's atomic dq plan) must agree with the default split plan —
+    all accumulate fp32; the gaps are summation order, the P/dS scratch
+    round trips, and the split plan's augmentation-fold constants
+    (lse/delta folded as two-element T pairs — T-squared error), so the
+    bf16 bound is wider than the fp16 one."""
     b, s, n, d = 1, 768, 3, 128
     q = _rand(b, s, n, d, dtype, 101)
     k = _rand(b, s, n, d, dtype, 102)
@@ -328,10 +331,14 @@ def test_bwd_fused_vs_split(causal, dtype):
     scale = 1.0 / math.sqrt(d)
     o, lse = ext.attn_fwd(q, k, v, scale, causal)
     delta = ext.attn_bwd_preprocess(o.to(dtype), do)
-    fused = ext.attn_bwd(do, q, k, v, delta, lse, scale, causal, False)
+    monkeypatch.setenv("BA_BWD_FUSED", plan)
+    alt = ext.attn_bwd(do, q, k, v, delta, lse, scale, causal, False)
+    monkeypatch.setenv("BA_BWD_FUSED", "0")
     split = ext.attn_bwd(do, q, k, v, delta, lse, scale, causal, True)
-    for f, s_, name in zip(fused, split, ("dq", "dk", "dv")):
-        torch.testing.assert_close(f, s_, rtol=2e-3, atol=1e-3, msg=name)
+    tol = (dict(rtol=2e-3, atol=1e-3) if dtype == torch.float16
+           else dict(rtol=1e-2, atol=8e-3))
+    for f, s_, name in zip(alt, split, ("dq", "dk", "dv")):
+        torch.testing.assert_close(f, s_, msg=name, **tol)
 
 
 @pytest.mark.parametrize("det", [False, True])
