@@ -39,7 +39,12 @@ namespace {
 // VPATH: 0 = V transposed image + b128 row-slice reads;
 //         1 = V row-major (tr16-swizzled) + ds_read_tr16_b64 fragments
 // SUBT: 0 = joint softmax over the 64-kv tile; 1 = per-32-subtile online
-// updates (lets subtile-0 PV MFMAs overlap subtile-1 QK/softmax)
+// updates (lets subtile-0 PV MFMAs overlap subtile-1 QK/softmax);
+// 2 = att[2] double-pipeline (guide T15): the softmax FINISH (mask, max,
+// rescale, exp2, pack) and PV of subtile j-1 are issued while subtile
+// j's QK MFMAs fill — an explicit two-stage software pipeline carried
+// ACROSS tile boundaries (requires NBUF=4 so the previous tile's V
+// image survives one extra subtile, and a barrier every tile)
 // NT: threads per workgroup (512 = 8 waves x 1 block/CU;
 //     256 = 4 waves x 2 blocks/CU — decoupled barrier groups)
 // NBUF: LDS tile buffers. 2 = stage one tile ahead, barrier every tile.
@@ -124,6 +129,59 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
   const int kv_limit =
       causal ? min(Sk, (int)(blockIdx.x + 1) * QROWS) : Sk;
   const int nt = (kv_limit + KVBLK - 1) / KVBLK;
+  static_assert(SUBT != 2 || (NBUF == 4 && VPATH == 0),
+                "the T15 pipeline needs NBUF=4 and the V^T image");
+
+  // ---- SUBT=2 pipeline state: the previous subtile's raw scores and
+  // its metadata, finished during the next subtile's QK cluster
+  f32x16_t stP;
+  int p_kv0 = -1;   // -1 = pipeline empty
+  int p_cur = 0;
+  bool p_full = false;
+  constexpr float DEFER_THR2 = 8.f;
+  auto finish_subtile = [&]() {
+    if (!p_full) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int kv_g = p_kv0 + ba_crow(r, 0) + 4 * hi;
+        if (!(kv_g < Sk && (!causal || kv_g <= q_row))) stP[r] = BA_NEG_BIG;
+      }
+    }
+    float tm = ba_max16(stP);
+    tm = fmaxf(tm, __shfl_xor(tm, 32));
+    tm *= c2;
+    if (!__all(tm - m2 <= DEFER_THR2)) {
+      const float mnew = fmaxf(m2, tm);
+      const float alpha = ba_exp2(m2 - mnew);
+      m2 = mnew;
+      lsum *= alpha;
+#pragma unroll
+      for (int dt = 0; dt < D / 32; ++dt)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) ot[dt][r] *= alpha;
+    }
+    float rowsum = 0.f;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      stP[r] = ba_exp2(__builtin_fmaf(stP[r], c2, -m2));
+      rowsum += stP[r];
+    }
+    rowsum += __shfl_xor(rowsum, 32);
+    lsum += rowsum;
+    frag pfp[2];
+    ba_build_frag_pair<T>(stP, pfp);
+    const int kvs_l = (p_kv0 / 32) & 1;  // subtile index within its tile
+#pragma unroll
+    for (int dt = 0; dt < D / 32; ++dt) {
+      const int drow = dt * 32 + l31;
+#pragma unroll
+      for (int u = 0; u < 2; ++u) {
+        frag vv = ba_ld_rowslice<T, KVBLK, SWZ_V, 7>(
+            ldsVT(p_cur), drow, kvs_l * 32 + 16 * u + 8 * hi);
+        ot[dt] = MT::mma(vv, pfp[u], ot[dt]);
+      }
+    }
+  };
 
   auto issue_loads = [&](int tile, u32x4_t* kreg, u32x4_t* vreg) {
     const int kv0 = tile * KVBLK;
@@ -181,7 +239,26 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
     if (has_next) issue_loads(t + AHEAD, kreg, vreg);
 
     const bool active = !causal || (kv0 <= qb + 31);
-    if (active && SUBT == 1) {
+    if (active && SUBT == 2) {
+      // ---- T15 pipeline: QK(j) fills while FINISH+PV(j-1) retire
+      const bool tile_full =
+          (kv0 + KVBLK <= Sk) && (!causal || (kv0 + KVBLK - 1 <= qb));
+#pragma unroll
+      for (int kvs = 0; kvs < 2; ++kvs) {
+        f32x16_t stQ = (f32x16_t)(0.f);
+#pragma unroll
+        for (int s2 = 0; s2 < D / 16; ++s2) {
+          frag kf = ba_ld_rowslice<T, D, SWZ_K>(ldsK(cur), kvs * 32 + l31,
+                                                16 * s2 + 8 * hi);
+          stQ = MT::mma(kf, qf[s2], stQ);
+        }
+        if (p_kv0 >= 0) finish_subtile();
+        stP = stQ;
+        p_kv0 = kv0 + kvs * 32;
+        p_cur = cur;
+        p_full = tile_full;
+      }
+    } else if (active && SUBT == 1) {
       // ---- per-subtile pipeline: {QK, softmax, PV} x 2, independent
       // chains so the scheduler overlaps PV(0) with QK(1)
       const bool tile_full =
@@ -334,8 +411,11 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
     }
 
     if (has_next) write_lds((t + AHEAD) % NBUF, kreg, vreg);
-    if (NBUF == 2 || (t & 1) || t + 1 >= nt) __syncthreads();
+    // SUBT=2 reads the previous tile's V one subtile late: barrier every
+    // tile so the rewrite (2 buffers ahead) never crosses those reads
+    if (SUBT == 2 || NBUF == 2 || (t & 1) || t + 1 >= nt) __syncthreads();
   }
+  if (SUBT == 2 && p_kv0 >= 0) finish_subtile();  // drain the pipeline
 
   // ---- epilogue
   if (q_row < Sq) {
@@ -462,6 +542,17 @@ static int launch_fwd(const void* q, const void* k, const void* v, float* o,
     const char* e = getenv("BA_FWD_NBUF");
     return e ? atoi(e) : 2;
   }();
+  if (subt == 2) {
+    attn_fwd_kernel<T, D, 64, 0, 0, 2, 512, 4>
+        <<<dim3((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B),
+           dim3(512), 0, (hipStream_t)stream>>>(
+            (const T*)q, (const T*)k, (const T*)v, o, lse, (int)Sq, (int)Sk,
+            (int)N, qs[0], qs[1], qs[2], ks[0], ks[1], ks[2], vs[0], vs[1],
+            vs[2], scale, causal, nullptr, nullptr, nullptr, 0, 0, 0, 0, 0,
+            0);
+    BA_CHECK_LAUNCH();
+    return 0;
+  }
   if (nbuf == 4 && ntw == 512) {
     if (vpath == 0)
       attn_fwd_kernel<T, D, 64, 0, 0, 1, 512, 4>
@@ -526,6 +617,17 @@ static int launch_fwd_accum(const void* q, const void* k, const void* v,
     const char* e = getenv("BA_FWD_NBUF");
     return e ? atoi(e) : 2;
   }();
+  if (subt == 2) {
+    dim3 grid2((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B);
+    attn_fwd_kernel<T, D, 64, 1, 0, 2, 512, 4>
+        <<<grid2, dim3(512), 0, (hipStream_t)stream>>>(
+            (const T*)q, (const T*)k, (const T*)v, nullptr, nullptr, (int)Sq,
+            (int)Sk, (int)N, qs[0], qs[1], qs[2], ks[0], ks[1], ks[2], vs[0],
+            vs[1], vs[2], scale, causal, acc, m, l, as[0], as[1], as[2],
+            mls[0], mls[1], carry_in);
+    BA_CHECK_LAUNCH();
+    return 0;
+  }
   if (nbuf == 4 && ntw == 512) {
     dim3 grid4((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B);
     if (vpath == 0)
