@@ -106,9 +106,51 @@ def bench_single_flash(args, b, s, n, d, causal, steps, warmup):
     }
 
 
+def bench_ring_naive(args, b, s_global, n, d, steps, warmup):
+    """RingQK/RingAV comparator column (reference benchmarks/ring_attn.py;
+    restated in benchmarks/ring_naive.py).  Materialises the full score
+    slab — configs must fit O(S^2/W) per rank."""
+    from benchmarks.ring_naive import ring_naive_attention, slab_bytes
+
+    world = dist.get_world_size()
+    rank = dist.get_rank()
+    s_local = s_global // world
+    dtype = torch.float16 if args.dtype == "fp16" else torch.bfloat16
+    need = 4 * slab_bytes(b, n, s_local, s_global, dtype)  # slab+probs+grads
+    free = torch.cuda.mem_get_info()[0]
+    if need > free * 0.6:
+        return None
+    g = torch.Generator().manual_seed(2000 + rank)
+    mk = lambda: torch.randn(b * n, s_local, d, generator=g).to(dtype).cuda()
+    q, k, v, do = mk(), mk(), mk(), mk()
+    scale = 1.0 / math.sqrt(d)
+
+    def fwd():
+        with torch.no_grad():
+            ring_naive_attention(q, k, v, scale)
+
+    def fwdbwd():
+        qg, kg, vg = (t.detach().requires_grad_() for t in (q, k, v))
+        o = ring_naive_attention(qg, kg, vg, scale)
+        torch.autograd.grad(o, (qg, kg, vg), do)
+
+    t_f = timeit(fwd, steps, warmup)
+    t_fb = timeit(fwdbwd, steps, warmup)
+    f = flops_fwd(b, s_global, n, d, False)
+    return {
+        "method": "ring_naive", "b": b, "s": s_global, "n": n, "d": d,
+        "causal": False, "wsize": world, "dtype": args.dtype,
+        "fwd_ms": round(t_f * 1e3, 2),
+        "fwd_tflops_per_gpu": round(f / t_f / 1e12 / world, 2),
+        "fwdbwd_ms": round(t_fb * 1e3, 2),
+        "fwdbwd_tflops_per_gpu": round(3.5 * f / t_fb / 1e12 / world, 2),
+    }
+
+
 def main():
     p = argparse.ArgumentParser()
-    p.add_argument("--configs", choices=["quick", "ring", "batch", "single"],
+    p.add_argument("--configs",
+                   choices=["quick", "ring", "batch", "single", "naive"],
                    default="quick")
     p.add_argument("--steps", type=int, default=3)
     p.add_argument("--warmup", type=int, default=1)
@@ -150,6 +192,14 @@ def main():
         for s in (65536, 131072, 262144):
             rows.append(bench_single_flash(args, 1, s, n, d, False,
                                            args.steps, args.warmup))
+    elif args.configs == "naive":  # RingQK/RingAV comparator column
+        for s in (8192, 16384, 32768):
+            r = bench_ring_naive(args, 1, s, n, d, args.steps, args.warmup)
+            if r is not None:
+                rows.append(r)
+            r = bench_burst(args, 1, s, n, d, False, False, False,
+                            args.steps, args.warmup)
+            rows.append(r)
 
     if rank == 0:
         with open(args.out, "a") as f:
