@@ -52,8 +52,14 @@ namespace {
 // rewritten two tiles after its last read, so one barrier in any two
 // consecutive tile boundaries separates every write from its readers
 // and every reader from the overwrite).
+// KREG: 1 = K tiles go straight from HBM into the MFMA fragment
+// registers (no LDS round trip): the QK cluster is PURE-REG (guide T16's
+// load-K->reg cluster), and tile t+1's K loads are issued the moment
+// tile t's QK MFMAs consume the block (WAR on the same registers), so
+// softmax+PV+staging+barrier (~700+ cycles) cover the HBM latency.
+// Only the V^T image stays in LDS.  Requires SUBT=1, NBUF=2, VPATH=0.
 template <typename T, int D, int KVBLK, int OUT_STATE, int VPATH, int SUBT,
-          int NT = 512, int NBUF = 2>
+          int NT = 512, int NBUF = 2, int KREG = 0>
 __global__ __launch_bounds__(NT) void attn_fwd_kernel(
     const T* __restrict__ q, const T* __restrict__ k, const T* __restrict__ v,
     float* __restrict__ o, float* __restrict__ lse,
@@ -75,11 +81,15 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
   constexpr int QROWS = NT / 2;  // q rows per workgroup (32 per wave)
   static_assert(PT >= 1, "tile must fill at least one chunk per thread");
 
+  static_assert(!KREG || (SUBT == 1 && NBUF == 2 && VPATH == 0),
+                "KREG path: SUBT=1, NBUF=2, V^T image only");
   // single LDS object: [NBUF buffers][K row-major | V transposed][KVBLK*D]
-  __shared__ T lds[NBUF * 2 * KVBLK * D];
-  auto ldsK = [&](int buf) -> T* { return lds + buf * (2 * KVBLK * D); };
+  // (KREG: V transposed only)
+  constexpr int IMGS_F = KREG ? 1 : 2;
+  __shared__ T lds[NBUF * IMGS_F * KVBLK * D];
+  auto ldsK = [&](int buf) -> T* { return lds + buf * (IMGS_F * KVBLK * D); };
   auto ldsVT = [&](int buf) -> T* {
-    return lds + buf * (2 * KVBLK * D) + KVBLK * D;
+    return lds + buf * (IMGS_F * KVBLK * D) + (KREG ? 0 : KVBLK * D);
   };
 
   const int tid = threadIdx.x;
@@ -196,7 +206,8 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
       // nullified by the kv-range mask (p = 0) in the softmax
       const int kvg = kv0 + row;
       const int kvc = kvg < Sk ? kvg : (Sk - 1);
-      kreg[c] = *(const u32x4_t*)(kp + (int64_t)kvc * k_ss + col8 * 8);
+      if (!KREG)
+        kreg[c] = *(const u32x4_t*)(kp + (int64_t)kvc * k_ss + col8 * 8);
       vreg[c] = *(const u32x4_t*)(vp + (int64_t)kvc * v_ss + col8 * 8);
     }
   };
@@ -206,19 +217,37 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
       const int flat = tid + c * NT;
       const int row = flat / (D / 8);
       const int col8 = flat % (D / 8);
-      const int byte = ba_swz<SWZ_K>(row * (2 * D) + col8 * 16, row);
-      *(u32x4_t*)((char*)ldsK(buf) + byte) = kreg[c];
+      if (!KREG) {
+        const int byte = ba_swz<SWZ_K>(row * (2 * D) + col8 * 16, row);
+        *(u32x4_t*)((char*)ldsK(buf) + byte) = kreg[c];
+      }
       if (VPATH == 0)
         ba_st_transposed<T, KVBLK, SWZ_V, 7>(ldsVT(buf), row, col8 * 8, vreg[c]);
       else
         ba_st_tr16row<T, D>(ldsVT(buf), row, col8 * 8, vreg[c]);
     }
   };
+  // KREG: one K subtile straight into the A-fragment registers (lane
+  // reads its own kv row; clamped like the staging)
+  auto load_k_sub = [&](int tile, int kvs, frag* dst) {
+#pragma unroll
+    for (int s = 0; s < D / 16; ++s) {
+      const int kvg = tile * KVBLK + kvs * 32 + l31;
+      const int kvc = kvg < Sk ? kvg : (Sk - 1);
+      dst[s] = __builtin_bit_cast(
+          frag, *(const u32x4_t*)(kp + (int64_t)kvc * k_ss + 16 * s + 8 * hi));
+    }
+  };
 
+  frag kfr[KREG ? 2 : 1][D / 16];  // KREG: the resident K tile fragments
   {  // prologue: tiles 0..NBUF-2
     u32x4_t kreg[PT], vreg[PT];
     issue_loads(0, kreg, vreg);
     write_lds(0, kreg, vreg);
+    if (KREG) {
+      load_k_sub(0, 0, kfr[0]);
+      load_k_sub(0, 1, kfr[KREG ? 1 : 0]);
+    }
     if (NBUF == 4 && nt > 1) {
       issue_loads(1, kreg, vreg);
       write_lds(1, kreg, vreg);
@@ -269,10 +298,15 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
         f32x16_t st = (f32x16_t)(0.f);
 #pragma unroll
         for (int s2 = 0; s2 < D / 16; ++s2) {
-          frag kf = ba_ld_rowslice<T, D, SWZ_K>(ldsK(cur), kvs * 32 + l31,
-                                                16 * s2 + 8 * hi);
+          frag kf = KREG ? kfr[KREG ? kvs : 0][s2]
+                         : ba_ld_rowslice<T, D, SWZ_K>(
+                               ldsK(cur), kvs * 32 + l31, 16 * s2 + 8 * hi);
           st = MT::mma(kf, qf[s2], st);
         }
+        // KREG: this subtile's K block is consumed — re-issue it for the
+        // next tile at once (WAR on the same registers; the HBM latency
+        // hides under softmax + PV + staging + the barrier)
+        if (KREG && has_next) load_k_sub(t + AHEAD, kvs, kfr[KREG ? kvs : 0]);
         // fold the softmax scale into the exp argument (exp2+fma): the
         // row max is taken on RAW scores (max commutes with c2 > 0), the
         // scale costs one multiply on the max instead of 16 per subtile
@@ -542,6 +576,21 @@ static int launch_fwd(const void* q, const void* k, const void* v, float* o,
     const char* e = getenv("BA_FWD_NBUF");
     return e ? atoi(e) : 2;
   }();
+  static const int kreg = [] {
+    const char* e = getenv("BA_FWD_KREG");
+    return e ? atoi(e) : 0;
+  }();
+  if (kreg == 1) {
+    attn_fwd_kernel<T, D, 64, 0, 0, 1, 512, 2, 1>
+        <<<dim3((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B),
+           dim3(512), 0, (hipStream_t)stream>>>(
+            (const T*)q, (const T*)k, (const T*)v, o, lse, (int)Sq, (int)Sk,
+            (int)N, qs[0], qs[1], qs[2], ks[0], ks[1], ks[2], vs[0], vs[1],
+            vs[2], scale, causal, nullptr, nullptr, nullptr, 0, 0, 0, 0, 0,
+            0);
+    BA_CHECK_LAUNCH();
+    return 0;
+  }
   if (subt == 2) {
     attn_fwd_kernel<T, D, 64, 0, 0, 2, 512, 4>
         <<<dim3((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B),
@@ -617,6 +666,21 @@ static int launch_fwd_accum(const void* q, const void* k, const void* v,
     const char* e = getenv("BA_FWD_NBUF");
     return e ? atoi(e) : 2;
   }();
+  static const int kreg = [] {
+    const char* e = getenv("BA_FWD_KREG");
+    return e ? atoi(e) : 0;
+  }();
+  if (kreg == 1) {
+    dim3 gridk((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B);
+    attn_fwd_kernel<T, D, 64, 1, 0, 1, 512, 2, 1>
+        <<<gridk, dim3(512), 0, (hipStream_t)stream>>>(
+            (const T*)q, (const T*)k, (const T*)v, nullptr, nullptr, (int)Sq,
+            (int)Sk, (int)N, qs[0], qs[1], qs[2], ks[0], ks[1], ks[2], vs[0],
+            vs[1], vs[2], scale, causal, acc, m, l, as[0], as[1], as[2],
+            mls[0], mls[1], carry_in);
+    BA_CHECK_LAUNCH();
+    return 0;
+  }
   if (subt == 2) {
     dim3 grid2((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B);
     attn_fwd_kernel<T, D, 64, 1, 0, 2, 512, 4>
