@@ -58,6 +58,13 @@ namespace {
 // tile t's QK MFMAs consume the block (WAR on the same registers), so
 // softmax+PV+staging+barrier (~700+ cycles) cover the HBM latency.
 // Only the V^T image stays in LDS.  Requires SUBT=1, NBUF=2, VPATH=0.
+// (Measured -47%: the per-lane strided K loads swamp vmem issue.)
+// KREG=2 = cross-barrier K-FRAGMENT PREFETCH: one 8-fragment register
+// set is re-filled from LDS a full SUBTILE ahead of its QK use — legal
+// across tile boundaries because NBUF=4 staging (AHEAD=2) wrote tile
+// t+1's buffer two tiles ago and a barrier every tile orders it.  The
+// QK cluster's lgkm park (SQ_WAIT_ANY 38% in the r2 PMC taxonomy)
+// becomes overlap.  Requires SUBT=1, NBUF=4, VPATH=0.
 template <typename T, int D, int KVBLK, int OUT_STATE, int VPATH, int SUBT,
           int NT = 512, int NBUF = 2, int KREG = 0>
 __global__ __launch_bounds__(NT) void attn_fwd_kernel(
@@ -81,15 +88,17 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
   constexpr int QROWS = NT / 2;  // q rows per workgroup (32 per wave)
   static_assert(PT >= 1, "tile must fill at least one chunk per thread");
 
-  static_assert(!KREG || (SUBT == 1 && NBUF == 2 && VPATH == 0),
-                "KREG path: SUBT=1, NBUF=2, V^T image only");
+  static_assert(KREG != 1 || (SUBT == 1 && NBUF == 2 && VPATH == 0),
+                "KREG=1 path: SUBT=1, NBUF=2, V^T image only");
+  static_assert(KREG != 2 || (SUBT == 1 && NBUF == 4 && VPATH == 0),
+                "KREG=2 path: SUBT=1, NBUF=4 (cross-barrier prefetch)");
   // single LDS object: [NBUF buffers][K row-major | V transposed][KVBLK*D]
-  // (KREG: V transposed only)
-  constexpr int IMGS_F = KREG ? 1 : 2;
+  // (KREG=1: V transposed only)
+  constexpr int IMGS_F = (KREG == 1) ? 1 : 2;
   __shared__ T lds[NBUF * IMGS_F * KVBLK * D];
   auto ldsK = [&](int buf) -> T* { return lds + buf * (IMGS_F * KVBLK * D); };
   auto ldsVT = [&](int buf) -> T* {
-    return lds + buf * (IMGS_F * KVBLK * D) + (KREG ? 0 : KVBLK * D);
+    return lds + buf * (IMGS_F * KVBLK * D) + (KREG == 1 ? 0 : KVBLK * D);
   };
 
   const int tid = threadIdx.x;
@@ -206,7 +215,7 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
       // nullified by the kv-range mask (p = 0) in the softmax
       const int kvg = kv0 + row;
       const int kvc = kvg < Sk ? kvg : (Sk - 1);
-      if (!KREG)
+      if (KREG != 1)
         kreg[c] = *(const u32x4_t*)(kp + (int64_t)kvc * k_ss + col8 * 8);
       vreg[c] = *(const u32x4_t*)(vp + (int64_t)kvc * v_ss + col8 * 8);
     }
@@ -217,7 +226,7 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
       const int flat = tid + c * NT;
       const int row = flat / (D / 8);
       const int col8 = flat % (D / 8);
-      if (!KREG) {
+      if (KREG != 1) {
         const int byte = ba_swz<SWZ_K>(row * (2 * D) + col8 * 16, row);
         *(u32x4_t*)((char*)ldsK(buf) + byte) = kreg[c];
       }
@@ -239,20 +248,26 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
     }
   };
 
-  frag kfr[KREG ? 2 : 1][D / 16];  // KREG: the resident K tile fragments
+  frag kfr[KREG == 1 ? 2 : 1][D / 16];  // KREG: resident K fragments
   {  // prologue: tiles 0..NBUF-2
     u32x4_t kreg[PT], vreg[PT];
     issue_loads(0, kreg, vreg);
     write_lds(0, kreg, vreg);
-    if (KREG) {
+    if (KREG == 1) {
       load_k_sub(0, 0, kfr[0]);
-      load_k_sub(0, 1, kfr[KREG ? 1 : 0]);
+      load_k_sub(0, 1, kfr[KREG == 1 ? 1 : 0]);
     }
     if (NBUF == 4 && nt > 1) {
       issue_loads(1, kreg, vreg);
       write_lds(1, kreg, vreg);
     }
     __syncthreads();
+    if (KREG == 2) {  // prime the prefetch pipeline: (tile 0, subtile 0)
+#pragma unroll
+      for (int s2 = 0; s2 < D / 16; ++s2)
+        kfr[0][s2] =
+            ba_ld_rowslice<T, D, SWZ_K>(ldsK(0), l31, 16 * s2 + 8 * hi);
+    }
   }
   // static priority for the younger dispatch half (T5 static form):
   // wave-uniform condition via readfirstlane, one s_setprio, no flips
@@ -298,15 +313,29 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
         f32x16_t st = (f32x16_t)(0.f);
 #pragma unroll
         for (int s2 = 0; s2 < D / 16; ++s2) {
-          frag kf = KREG ? kfr[KREG ? kvs : 0][s2]
+          frag kf = KREG ? kfr[KREG == 1 ? kvs : 0][s2]
                          : ba_ld_rowslice<T, D, SWZ_K>(
                                ldsK(cur), kvs * 32 + l31, 16 * s2 + 8 * hi);
           st = MT::mma(kf, qf[s2], st);
         }
-        // KREG: this subtile's K block is consumed — re-issue it for the
-        // next tile at once (WAR on the same registers; the HBM latency
-        // hides under softmax + PV + staging + the barrier)
-        if (KREG && has_next) load_k_sub(t + AHEAD, kvs, kfr[KREG ? kvs : 0]);
+        // KREG=1: this subtile's K block is consumed — re-issue its
+        // global loads for the next tile at once (WAR on the same
+        // registers; HBM latency hides under softmax + PV + staging)
+        if (KREG == 1 && has_next)
+          load_k_sub(t + AHEAD, kvs, kfr[KREG == 1 ? kvs : 0]);
+        // KREG=2: re-fill the fragment set from LDS one subtile ahead
+        // ((t,0)->(t,1)->(t+1,0); tile t+1's buffer was staged two
+        // tiles ago, ordered by the per-tile barrier)
+        if (KREG == 2) {
+          const int pn_t = (kvs == 0) ? t : t + 1;
+          if (pn_t < nt) {
+            const int pkvs = kvs ^ 1;
+#pragma unroll
+            for (int s2 = 0; s2 < D / 16; ++s2)
+              kfr[0][s2] = ba_ld_rowslice<T, D, SWZ_K>(
+                  ldsK(pn_t % NBUF), pkvs * 32 + l31, 16 * s2 + 8 * hi);
+          }
+        }
         // fold the softmax scale into the exp argument (exp2+fma): the
         // row max is taken on RAW scores (max commutes with c2 > 0), the
         // scale costs one multiply on the max instead of 16 per subtile
@@ -447,7 +476,8 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
     if (has_next) write_lds((t + AHEAD) % NBUF, kreg, vreg);
     // SUBT=2 reads the previous tile's V one subtile late: barrier every
     // tile so the rewrite (2 buffers ahead) never crosses those reads
-    if (SUBT == 2 || NBUF == 2 || (t & 1) || t + 1 >= nt) __syncthreads();
+    if (SUBT == 2 || KREG == 2 || NBUF == 2 || (t & 1) || t + 1 >= nt)
+      __syncthreads();
   }
   if (SUBT == 2 && p_kv0 >= 0) finish_subtile();  // drain the pipeline
 
@@ -580,14 +610,22 @@ static int launch_fwd(const void* q, const void* k, const void* v, float* o,
     const char* e = getenv("BA_FWD_KREG");
     return e ? atoi(e) : 0;
   }();
-  if (kreg == 1) {
-    attn_fwd_kernel<T, D, 64, 0, 0, 1, 512, 2, 1>
-        <<<dim3((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B),
-           dim3(512), 0, (hipStream_t)stream>>>(
-            (const T*)q, (const T*)k, (const T*)v, o, lse, (int)Sq, (int)Sk,
-            (int)N, qs[0], qs[1], qs[2], ks[0], ks[1], ks[2], vs[0], vs[1],
-            vs[2], scale, causal, nullptr, nullptr, nullptr, 0, 0, 0, 0, 0,
-            0);
+  if (kreg == 1 || kreg == 2) {
+    dim3 gk((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B);
+    if (kreg == 1)
+      attn_fwd_kernel<T, D, 64, 0, 0, 1, 512, 2, 1>
+          <<<gk, dim3(512), 0, (hipStream_t)stream>>>(
+              (const T*)q, (const T*)k, (const T*)v, o, lse, (int)Sq,
+              (int)Sk, (int)N, qs[0], qs[1], qs[2], ks[0], ks[1], ks[2],
+              vs[0], vs[1], vs[2], scale, causal, nullptr, nullptr, nullptr,
+              0, 0, 0, 0, 0, 0);
+    else
+      attn_fwd_kernel<T, D, 64, 0, 0, 1, 512, 4, 2>
+          <<<gk, dim3(512), 0, (hipStream_t)stream>>>(
+              (const T*)q, (const T*)k, (const T*)v, o, lse, (int)Sq,
+              (int)Sk, (int)N, qs[0], qs[1], qs[2], ks[0], ks[1], ks[2],
+              vs[0], vs[1], vs[2], scale, causal, nullptr, nullptr, nullptr,
+              0, 0, 0, 0, 0, 0);
     BA_CHECK_LAUNCH();
     return 0;
   }
@@ -670,14 +708,22 @@ static int launch_fwd_accum(const void* q, const void* k, const void* v,
     const char* e = getenv("BA_FWD_KREG");
     return e ? atoi(e) : 0;
   }();
-  if (kreg == 1) {
+  if (kreg == 1 || kreg == 2) {
     dim3 gridk((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B);
-    attn_fwd_kernel<T, D, 64, 1, 0, 1, 512, 2, 1>
-        <<<gridk, dim3(512), 0, (hipStream_t)stream>>>(
-            (const T*)q, (const T*)k, (const T*)v, nullptr, nullptr, (int)Sq,
-            (int)Sk, (int)N, qs[0], qs[1], qs[2], ks[0], ks[1], ks[2], vs[0],
-            vs[1], vs[2], scale, causal, acc, m, l, as[0], as[1], as[2],
-            mls[0], mls[1], carry_in);
+    if (kreg == 1)
+      attn_fwd_kernel<T, D, 64, 1, 0, 1, 512, 2, 1>
+          <<<gridk, dim3(512), 0, (hipStream_t)stream>>>(
+              (const T*)q, (const T*)k, (const T*)v, nullptr, nullptr,
+              (int)Sq, (int)Sk, (int)N, qs[0], qs[1], qs[2], ks[0], ks[1],
+              ks[2], vs[0], vs[1], vs[2], scale, causal, acc, m, l, as[0],
+              as[1], as[2], mls[0], mls[1], carry_in);
+    else
+      attn_fwd_kernel<T, D, 64, 1, 0, 1, 512, 4, 2>
+          <<<gridk, dim3(512), 0, (hipStream_t)stream>>>(
+              (const T*)q, (const T*)k, (const T*)v, nullptr, nullptr,
+              (int)Sq, (int)Sk, (int)N, qs[0], qs[1], qs[2], ks[0], ks[1],
+              ks[2], vs[0], vs[1], vs[2], scale, causal, acc, m, l, as[0],
+              as[1], as[2], mls[0], mls[1], carry_in);
     BA_CHECK_LAUNCH();
     return 0;
   }
