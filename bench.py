@@ -229,7 +229,8 @@ def main():
     # 1 - exposed/comm_only, where exposed = full - compute_only is the
     # comm time NOT hidden under the tile kernels (north_star: >=90%).
     ring_stats = None
-    if world > 1:
+    try:
+      if world > 1:
         from burst_attn_amd.comm import Ring
         from burst_attn_amd.tile import get_tile_provider
 
@@ -261,6 +262,9 @@ def main():
             "overlap_frac": round(max(0.0, min(1.0, 1.0 - exposed / t_comm)), 4)
             if t_comm > 0 else None,
         }
+    except Exception as e:  # never let the overlap probe kill the bench
+        log(f"[bench] ring overlap probe failed: {e!r}")
+        ring_stats = {"error": repr(e)}
 
     cpu_baseline = None
     if rank == 0 and world == 1 and not args.no_cpu_baseline:
