@@ -352,14 +352,17 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
       greg[c] = *(const u32x4_t*)(gp + (int64_t)qc * g_ss + col8 * 8);
     }
     // lse*log2e and delta rows, interleaved {lse2, delta} per q row so
-    // the compute loop reads one 8B word per row
-    lsed[0] = 0.f;
-    if (tid < QBLK) {
-      const int qg = q0 + tid;
-      lsed[0] = (qg < Sq) ? lp_[qg] * BA_LOG2E : 0.f;
-    } else if (tid < 2 * QBLK && MODE >= 1) {
-      const int qg = q0 + tid - QBLK;
-      lsed[0] = (qg < Sq) ? dp_[qg] : 0.f;
+    // the compute loop reads one 8B word per row.  Loads are CLAMPED and
+    // UNCONDITIONAL (a guarded load puts an exec branch around the whole
+    // staging batch and serialises it — the same lesson as the q/dO
+    // clamps above); only the select below is masked.
+    {
+      const int idx = tid & (QBLK - 1);
+      const int qg = q0 + idx;
+      const int qc = qg < Sq ? qg : (Sq - 1);
+      const float lv = lp_[qc] * BA_LOG2E;
+      const float dv_ = (MODE >= 1) ? dp_[qc] : 0.f;
+      lsed[0] = (qg < Sq) ? ((tid < QBLK) ? lv : dv_) : 0.f;
     }
   };
   auto write_lds = [&](int buf, const u32x4_t* qreg, const u32x4_t* greg,
@@ -405,6 +408,12 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
 
     const bool active = !causal || (q0 + QBLK - 1 >= kvb);
     if (active) {
+      // hoist the per-subtile {lse2, delta} LDS reads to tile start so
+      // their latency settles before the first aug MFMA needs them
+      float2 ld2s[QBLK / 32];
+#pragma unroll
+      for (int qs = 0; qs < QBLK / 32; ++qs)
+        ld2s[qs] = *(const float2*)&ldsF(cur)[2 * (qs * 32 + l31)];
 #pragma unroll
       for (int qs = 0; qs < QBLK / 32; ++qs) {
         // ---- augmentation fold: the per-row constants ride the MFMA.
@@ -415,7 +424,7 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
         // argument is just st*c2; dP accumulates -delta so dS is just
         // p*dpt*scale.  Replaces 32 per-row LDS reads + fma/sub chains
         // (68% WAIT_ANY and 24 spilled VGPRs on the dK kernel).
-        const float2 ld2 = *(const float2*)&ldsF(cur)[2 * (qs * 32 + l31)];
+        const float2 ld2 = ld2s[qs];
         frag ones01, ones0, a_lse, a_dlt;
         {
           u32x4_t z = {0, 0, 0, 0};
