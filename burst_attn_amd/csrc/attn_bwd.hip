@@ -270,8 +270,13 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
   constexpr int IMGS = MODE == 0 ? 2 : (MODE == 1 ? 3 : 4);
 
   // one LDS object (a second __shared__ forces vmcnt(0) on every ds_read
-  // — cdna guide §5 trap 4a): tail carved for per-tile lse2/delta floats
-  __shared__ T lds[2 * IMGS * QBLK * D + 2 * (2 * QBLK) * (4 / (int)sizeof(T))];
+  // — cdna guide §5 trap 4a).  lse/delta do NOT go through LDS: a
+  // float-cast tail here defeated alias analysis and made hipcc emit
+  // lgkmcnt(0) before EVERY ds_read of the stream images (50 full
+  // drains per tile vs the dq kernel's counted waits — the round-2 wait
+  // taxonomy's 58% parked dK).  They ride per-lane global loads,
+  // prefetched one tile ahead, instead.
+  __shared__ T lds[2 * IMGS * QBLK * D];
   auto ldsQ = [&](int buf) -> T* { return lds + buf * (IMGS * QBLK * D); };
   // MODE_DV: transposed dO; MODE_DK: row-major dO
   auto ldsG = [&](int buf) -> T* {
@@ -282,10 +287,6 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
   };
   auto ldsGT2 = [&](int buf) -> T* {  // MODE 2 only: dO^T as 4th image
     return lds + buf * (IMGS * QBLK * D) + 3 * QBLK * D;
-  };
-  // per-buffer float tail: [0..QBLK) = lse * log2e, [QBLK..2*QBLK) = delta
-  auto ldsF = [&](int buf) -> float* {
-    return (float*)(lds + 2 * IMGS * QBLK * D) + buf * (2 * QBLK);
   };
 
   const int tid = threadIdx.x;
@@ -337,7 +338,10 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
   const int t0 = causal ? (blockIdx.x * 256) / QBLK : 0;
   const int nt = (Sq + QBLK - 1) / QBLK;
 
-  auto issue_loads = [&](int tile, u32x4_t* qreg, u32x4_t* greg, float* lsed) {
+  // per-lane lse*log2e and delta for both subtiles of a tile (l2[qs] =
+  // {lse2, delta}), loaded straight from global — clamped, branchless
+  auto issue_loads = [&](int tile, u32x4_t* qreg, u32x4_t* greg,
+                         float2* l2) {
     const int q0 = tile * QBLK;
 #pragma unroll
     for (int c = 0; c < PT; ++c) {
@@ -351,24 +355,16 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
       qreg[c] = *(const u32x4_t*)(qp + (int64_t)qc * q_ss + col8 * 8);
       greg[c] = *(const u32x4_t*)(gp + (int64_t)qc * g_ss + col8 * 8);
     }
-    // lse*log2e and delta rows, interleaved {lse2, delta} per q row so
-    // the compute loop reads one 8B word per row.  Loads are CLAMPED and
-    // UNCONDITIONAL (a guarded load puts an exec branch around the whole
-    // staging batch and serialises it — the same lesson as the q/dO
-    // clamps above); only the select below is masked.
-    {
-      const int idx = tid & (QBLK - 1);
-      const int qg = q0 + idx;
+#pragma unroll
+    for (int qs = 0; qs < QBLK / 32; ++qs) {
+      const int qg = q0 + qs * 32 + l31;
       const int qc = qg < Sq ? qg : (Sq - 1);
-      const float lv = lp_[qc] * BA_LOG2E;
-      const float dv_ = (MODE >= 1) ? dp_[qc] : 0.f;
-      lsed[0] = (qg < Sq) ? ((tid < QBLK) ? lv : dv_) : 0.f;
+      // clamped rows contaminate nothing: the q_g < Sq mask zeroes p/dS
+      l2[qs].x = lp_[qc] * BA_LOG2E;
+      l2[qs].y = (MODE >= 1) ? dp_[qc] : 0.f;
     }
   };
-  auto write_lds = [&](int buf, const u32x4_t* qreg, const u32x4_t* greg,
-                       float lsed) {
-    if (tid < QBLK) ldsF(buf)[2 * tid] = lsed;
-    else if (tid < 2 * QBLK) ldsF(buf)[2 * (tid - QBLK) + 1] = lsed;
+  auto write_lds = [&](int buf, const u32x4_t* qreg, const u32x4_t* greg) {
 #pragma unroll
     for (int c = 0; c < PT; ++c) {
       const int flat = tid + c * NT;
@@ -388,11 +384,11 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
     }
   };
 
+  float2 l2c[QBLK / 32];
   {
     u32x4_t qreg[PT], greg[PT];
-    float lsed;
-    issue_loads(t0, qreg, greg, &lsed);
-    write_lds(0, qreg, greg, lsed);
+    issue_loads(t0, qreg, greg, l2c);
+    write_lds(0, qreg, greg);
     __syncthreads();
   }
   if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
@@ -403,17 +399,12 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
     const int q0 = t * QBLK;
     const bool has_next = (t + 1) < nt;
     u32x4_t qreg[PT], greg[PT];
-    float lsed;
-    if (has_next) issue_loads(t + 1, qreg, greg, &lsed);
+    float2 l2n[QBLK / 32] = {};
+    if (has_next) issue_loads(t + 1, qreg, greg, l2n);
 
     const bool active = !causal || (q0 + QBLK - 1 >= kvb);
     if (active) {
-      // hoist the per-subtile {lse2, delta} LDS reads to tile start so
-      // their latency settles before the first aug MFMA needs them
-      float2 ld2s[QBLK / 32];
-#pragma unroll
-      for (int qs = 0; qs < QBLK / 32; ++qs)
-        ld2s[qs] = *(const float2*)&ldsF(cur)[2 * (qs * 32 + l31)];
+      const float2* ld2s = l2c;  // this tile's lse2/delta (prefetched)
 #pragma unroll
       for (int qs = 0; qs < QBLK / 32; ++qs) {
         // ---- augmentation fold: the per-row constants ride the MFMA.
@@ -501,9 +492,11 @@ __global__ __launch_bounds__(512) void bwd_dkv_kernel(
         }
       }
     }
-    if (has_next) write_lds(cur ^ 1, qreg, greg, lsed);
+    if (has_next) write_lds(cur ^ 1, qreg, greg);
     __syncthreads();
     cur ^= 1;
+#pragma unroll
+    for (int qs = 0; qs < QBLK / 32; ++qs) l2c[qs] = l2n[qs];
   }
 
   if (kv_col < Sk) {
