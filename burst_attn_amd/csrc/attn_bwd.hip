@@ -73,7 +73,12 @@ __global__ __launch_bounds__(256) void bwd_preprocess_kernel(
 // ====================== dq kernel (q-resident) ========================
 constexpr int KVBLK = 64;
 
-template <typename T, int D>
+// DQP=1 (env BA_DQ_PIPE): register-relief + explicit read-ahead
+// experiment driven by the r2 wait-taxonomy finding: the
+// dO fragments move from 32 registers to a persistent LDS image, and
+// the freed registers hold a 2-deep software pipeline of the k/v
+// fragment reads so each ds_read has >= 2 MFMAs of cover.
+template <typename T, int D, int DQP = 0>
 __global__ __launch_bounds__(512) void bwd_dq_kernel(
     const T* __restrict__ dout, const T* __restrict__ q,
     const T* __restrict__ k, const T* __restrict__ v,
@@ -94,10 +99,12 @@ __global__ __launch_bounds__(512) void bwd_dq_kernel(
   constexpr int PT = (KVBLK * D / 8) / NT;
 
   // [2 buffers][K row-major | V row-major | K^T transposed]
-  __shared__ T lds[2 * 3 * KVBLK * D];
+  // (+ DQP: persistent dO image for the workgroup's 256 q rows)
+  __shared__ T lds[2 * 3 * KVBLK * D + (DQP ? 256 * D : 0)];
   auto ldsK = [&](int buf) -> T* { return lds + buf * (3 * KVBLK * D); };
   auto ldsV = [&](int buf) -> T* { return lds + buf * (3 * KVBLK * D) + KVBLK * D; };
   auto ldsKT = [&](int buf) -> T* { return lds + buf * (3 * KVBLK * D) + 2 * KVBLK * D; };
+  T* ldsDO = lds + 2 * 3 * KVBLK * D;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -112,18 +119,35 @@ __global__ __launch_bounds__(512) void bwd_dq_kernel(
   const T* kp = k + b * k_sb + (int64_t)n * k_sh;
   const T* vp = v + b * v_sb + (int64_t)n * v_sh;
 
-  frag qf[D / 16], gf[D / 16];
+  frag qf[D / 16], gf[DQP ? 1 : D / 16];
 #pragma unroll
   for (int s = 0; s < D / 16; ++s) {
     if (q_row < Sq) {
       qf[s] = __builtin_bit_cast(
           frag, *(const u32x4_t*)(qp + (int64_t)q_row * q_ss + 16 * s + 8 * hi));
-      gf[s] = __builtin_bit_cast(
-          frag, *(const u32x4_t*)(gp + (int64_t)q_row * g_ss + 16 * s + 8 * hi));
+      if (!DQP)
+        gf[s] = __builtin_bit_cast(
+            frag,
+            *(const u32x4_t*)(gp + (int64_t)q_row * g_ss + 16 * s + 8 * hi));
     } else {
       u32x4_t z = {0, 0, 0, 0};
       qf[s] = __builtin_bit_cast(frag, z);
-      gf[s] = __builtin_bit_cast(frag, z);
+      if (!DQP) gf[s] = __builtin_bit_cast(frag, z);
+    }
+  }
+  if (DQP) {
+    // one-time: stage the workgroup's dO block [256 q][D] row-major
+    // (clamped rows contribute garbage x 0 through the q_row mask)
+    constexpr int PTD = (256 * D / 8) / NT;
+#pragma unroll
+    for (int c = 0; c < PTD; ++c) {
+      const int flat = tid + c * NT;
+      const int row = flat / (D / 8), col8 = flat % (D / 8);
+      const int qg = blockIdx.x * 256 + row;
+      const int qc = qg < Sq ? qg : (Sq - 1);
+      u32x4_t ch = *(const u32x4_t*)(gp + (int64_t)qc * g_ss + col8 * 8);
+      const int byte = ba_swz<SWZ>(row * (2 * D) + col8 * 16, row);
+      *(u32x4_t*)((char*)ldsDO + byte) = ch;
     }
   }
   const float c2 = scale * BA_LOG2E;
@@ -186,6 +210,31 @@ __global__ __launch_bounds__(512) void bwd_dq_kernel(
 #pragma unroll
       for (int kvs = 0; kvs < 2; ++kvs) {
         f32x16_t st = (f32x16_t)(0.f), dpt = (f32x16_t)(0.f);
+        if (DQP) {
+          // explicit 2-deep read-ahead: fragment reads for slice s+1
+          // issue before slice s's MFMAs, so each ds_read is covered by
+          // >= 2 MFMAs instead of the allocator's read-wait-use
+          frag kfA = ba_ld_rowslice<T, D, SWZ>(ldsK(cur), kvs * 32 + l31,
+                                               8 * hi);
+          frag vfA = ba_ld_rowslice<T, D, SWZ>(ldsV(cur), kvs * 32 + l31,
+                                               8 * hi);
+          frag gfA = ba_ld_rowslice<T, D, SWZ>(ldsDO, wave * 32 + l31,
+                                               8 * hi);
+#pragma unroll
+          for (int s = 0; s < D / 16; ++s) {
+            frag kfB = kfA, vfB = vfA, gfB = gfA;
+            if (s + 1 < D / 16) {
+              kfA = ba_ld_rowslice<T, D, SWZ>(ldsK(cur), kvs * 32 + l31,
+                                              16 * (s + 1) + 8 * hi);
+              vfA = ba_ld_rowslice<T, D, SWZ>(ldsV(cur), kvs * 32 + l31,
+                                              16 * (s + 1) + 8 * hi);
+              gfA = ba_ld_rowslice<T, D, SWZ>(ldsDO, wave * 32 + l31,
+                                              16 * (s + 1) + 8 * hi);
+            }
+            st = MT::mma(kfB, qf[s], st);
+            dpt = MT::mma(vfB, gfB, dpt);
+          }
+        } else {
 #pragma unroll
         for (int s = 0; s < D / 16; ++s) {
           frag kf = ba_ld_rowslice<T, D, SWZ>(ldsK(cur), kvs * 32 + l31,
@@ -193,7 +242,8 @@ __global__ __launch_bounds__(512) void bwd_dq_kernel(
           frag vf = ba_ld_rowslice<T, D, SWZ>(ldsV(cur), kvs * 32 + l31,
                                               16 * s + 8 * hi);
           st = MT::mma(kf, qf[s], st);
-          dpt = MT::mma(vf, gf[s], dpt);
+          dpt = MT::mma(vf, gf[DQP ? 0 : s], dpt);
+        }
         }
         // p^T = exp2(S^T*c2 - lse2);  dS^T = p^T*(dP^T - delta)*scale
 #pragma unroll
@@ -1011,12 +1061,19 @@ static int launch_bwd(const void* dout, const void* q, const void* k,
       ls[1], dqs[0], dqs[1], dqs[2], dks[0], dks[1], dks[2], dvs[0], dvs[1],  \
       dvs[2], scale, causal
 
+  const char* dqp_e = getenv("BA_DQ_PIPE");
+  const int dqp = dqp_e ? atoi(dqp_e) : 0;
   if (plan != 2) {
-    bwd_dq_kernel<T, D><<<grid_dq, 512, 0, (hipStream_t)stream>>>(
-        (const T*)dout, (const T*)q, (const T*)k, (const T*)v, delta, lse, dq,
-        (int)Sq, (int)Sk, (int)N, gs[0], gs[1], gs[2], qs[0], qs[1], qs[2],
-        ks[0], ks[1], ks[2], vs[0], vs[1], vs[2], ds[0], ds[1], ls[0], ls[1],
-        dqs[0], dqs[1], dqs[2], scale, causal);
+#define DQ_ARGS                                                               \
+  (const T*)dout, (const T*)q, (const T*)k, (const T*)v, delta, lse, dq,      \
+      (int)Sq, (int)Sk, (int)N, gs[0], gs[1], gs[2], qs[0], qs[1], qs[2],     \
+      ks[0], ks[1], ks[2], vs[0], vs[1], vs[2], ds[0], ds[1], ls[0], ls[1],   \
+      dqs[0], dqs[1], dqs[2], scale, causal
+    if (dqp)
+      bwd_dq_kernel<T, D, 1><<<grid_dq, 512, 0, (hipStream_t)stream>>>(DQ_ARGS);
+    else
+      bwd_dq_kernel<T, D, 0><<<grid_dq, 512, 0, (hipStream_t)stream>>>(DQ_ARGS);
+#undef DQ_ARGS
     BA_CHECK_LAUNCH();
   }
   if (plan == 1) {

@@ -134,6 +134,12 @@ int main(int argc, char** argv) {
         s4[0], s4[1], s4[2], s4[0], s4[1], s4[2], s4[0], s4[1], s4[2], s2[0],
         s2[1], s2[0], s2[1], s4[0], s4[1], s4[2], scale, 0);
   });
+  bench("dq kernel DQP=1", [&] {
+    bwd_dq_kernel<T, 128, 1><<<gdq, 512>>>(
+        dout, q, k, v, delta, lse, dq, S, S, N, s4[0], s4[1], s4[2], s4[0],
+        s4[1], s4[2], s4[0], s4[1], s4[2], s4[0], s4[1], s4[2], s2[0], s2[1],
+        s2[0], s2[1], s4[0], s4[1], s4[2], scale, 0);
+  });
   bench("dq kernel", [&] {
     bwd_dq_kernel<T, 128><<<gdq, 512>>>(
         dout, q, k, v, delta, lse, dq, S, S, N, s4[0], s4[1], s4[2], s4[0],
