@@ -259,8 +259,10 @@ def test_accum_carry_in_matches_stateless(dtype):
     torch.testing.assert_close(o2[:, half:].float(), o_bot, **TOL[dtype])
 
 
-@pytest.mark.parametrize("opt_bwd,det", [(True, False), (False, True)])
-def test_interface_flags_single_rank(opt_bwd, det):
+@pytest.mark.parametrize("dtype", [torch.float16, torch.bfloat16])
+@pytest.mark.parametrize("opt_bwd,det", [(True, False), (False, True),
+                                         (True, True)])
+def test_interface_flags_single_rank(opt_bwd, det, dtype):
     """optimize_bwd_comm / deterministic flag combinations end to end on
     one GPU (the W>1 flavours are covered by the gloo ring tests)."""
     import torch.distributed as dist
@@ -274,7 +276,6 @@ def test_interface_flags_single_rank(opt_bwd, det):
     from burst_attn_amd import burst_attn_func
 
     b, s, n, d = 1, 512, 2, 128
-    dtype = torch.float16
     q = _rand(b, s, n, d, dtype, 81).requires_grad_()
     k = _rand(b, s, n, d, dtype, 82).requires_grad_()
     v = _rand(b, s, n, d, dtype, 83).requires_grad_()
@@ -284,10 +285,12 @@ def test_interface_flags_single_rank(opt_bwd, det):
     o_ref, dq_r, dk_r, dv_r = oracle.ring_forward_backward_reference(
         q.detach().cpu(), k.detach().cpu(), v.detach().cpu(), do.cpu(), None, True
     )
-    torch.testing.assert_close(o.float().cpu(), o_ref, rtol=2e-3, atol=1e-2)
-    torch.testing.assert_close(dq.float().cpu(), dq_r, rtol=5e-3, atol=2e-2)
-    torch.testing.assert_close(dk.float().cpu(), dk_r, rtol=5e-3, atol=2e-2)
-    torch.testing.assert_close(dv.float().cpu(), dv_r, rtol=5e-3, atol=2e-2)
+    otol = TOL[dtype]
+    btol = BWD_TOL[dtype]
+    torch.testing.assert_close(o.float().cpu(), o_ref, **otol)
+    torch.testing.assert_close(dq.float().cpu(), dq_r, **btol)
+    torch.testing.assert_close(dk.float().cpu(), dk_r, **btol)
+    torch.testing.assert_close(dv.float().cpu(), dv_r, **btol)
 
 
 def test_backward_bitwise_deterministic():
