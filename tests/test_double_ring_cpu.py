@@ -20,21 +20,22 @@ LOCAL = 2
 RTOL, ATOL = 1e-4, 1e-4
 
 
-def _make_groups(create_dq_group):
-    ranks = np.arange(WORLD).reshape(-1, LOCAL)
+def _make_groups(create_dq_group, world=WORLD, local=LOCAL):
+    ranks = np.arange(world).reshape(-1, local)
     intra = [dist.new_group(list(r)) for r in ranks]
     inter = [dist.new_group(list(r)) for r in ranks.T]
     me = dist.get_rank()
-    my_intra = intra[me // LOCAL]
-    my_inter = inter[me % LOCAL]
+    my_intra = intra[me // local]
+    my_inter = inter[me % local]
     if not create_dq_group:
         return my_intra, my_inter
     intra2 = [dist.new_group(list(r)) for r in ranks]
     inter2 = [dist.new_group(list(r)) for r in ranks.T]
-    return (my_intra, intra2[me // LOCAL]), (my_inter, inter2[me % LOCAL])
+    return (my_intra, intra2[me // local]), (my_inter, inter2[me % local])
 
 
-def _worker(rank, world, port, causal, striped, opt_bwd, dq_groups, fail_q):
+def _worker(rank, world, port, causal, striped, opt_bwd, dq_groups, fail_q,
+            local=LOCAL):
     try:
         os.environ["MASTER_ADDR"] = "127.0.0.1"
         os.environ["MASTER_PORT"] = str(port)
@@ -44,7 +45,7 @@ def _worker(rank, world, port, causal, striped, opt_bwd, dq_groups, fail_q):
         from .cpu_tile_provider import OracleTileProvider
 
         _set_tile_provider_for_testing(OracleTileProvider())
-        intra_g, inter_g = _make_groups(dq_groups)
+        intra_g, inter_g = _make_groups(dq_groups, world, local)
 
         b, s_local, n, d = 1, 64, 2, 32
         s = s_local * world
@@ -94,6 +95,28 @@ def test_double_ring_matches_full_attention(causal, striped, opt_bwd, dq_groups)
             _worker,
             args=(WORLD, _PORT[0], causal, striped, opt_bwd, dq_groups, fail_q),
             nprocs=WORLD,
+            join=True,
+        )
+    except Exception:
+        msgs = []
+        while not fail_q.empty():
+            msgs.append(fail_q.get())
+        raise AssertionError("double-ring test failed:\n" + "\n".join(msgs))
+
+
+@pytest.mark.parametrize("world,local", [(6, 3), (6, 2)])
+def test_double_ring_rectangular(world, local):
+    """Non-square hierarchies (3 "nodes" x 2 ranks and 2 x 3) — the
+    intra/inter phase arithmetic must hold when intra_size != inter_size
+    (reference comm.py:221-254 staging)."""
+    _PORT[0] += 1
+    ctx = mp.get_context("spawn")
+    fail_q = ctx.SimpleQueue()
+    try:
+        mp.spawn(
+            _worker,
+            args=(world, _PORT[0], True, False, False, True, fail_q, local),
+            nprocs=world,
             join=True,
         )
     except Exception:
