@@ -101,6 +101,29 @@ class Ring:
     ``Ring`` (``comm.py:104-321``) minus the bmtrain backend.
     """
 
+    # opt-in explicit side-stream commit (BA_RING_SIDE_STREAM=1): the
+    # fallback the reference builds with bmtrain side streams
+    # (comm.py:267-283) in case RCCL's internal-stream overlap measures
+    # short on a multi-GPU box (bench.py's `ring.overlap_frac`).  The
+    # commit is issued from a dedicated HIP stream that first waits on an
+    # event recorded on the compute stream, so the P2P enqueue decouples
+    # from the compute stream's queue; wait() still blocks the compute
+    # stream on the transfer.  Default OFF — RCCL's own streams carry the
+    # overlap in the default path.
+    _side_stream = None
+
+    @classmethod
+    def _maybe_side_stream(cls):
+        if os.environ.get("BA_RING_SIDE_STREAM", "0") != "1":
+            return None
+        import torch
+
+        if not torch.cuda.is_available():
+            return None
+        if cls._side_stream is None:
+            cls._side_stream = torch.cuda.Stream()
+        return cls._side_stream
+
     def __init__(self, process_group=None, double_group=(None, None)):
         dg = double_group if double_group is not None else (None, None)
         self.comm = process_group
@@ -240,12 +263,24 @@ class Ring:
         else:
             self.send_recv(tensor_list, dest_list, self.local_group)
 
+    def _commit_ops(self, ops):
+        side = self._maybe_side_stream()
+        if side is None:
+            return dist.batch_isend_irecv(ops)
+        import torch
+
+        ev = torch.cuda.Event()
+        ev.record()  # producers on the compute stream
+        with torch.cuda.stream(side):
+            side.wait_event(ev)
+            return dist.batch_isend_irecv(ops)
+
     def commit(self):
         if self.ops:
-            self.reqs += dist.batch_isend_irecv(self.ops)
+            self.reqs += self._commit_ops(self.ops)
             self.ops = []
         if self._inter_ops:
-            self._inter_reqs += dist.batch_isend_irecv(self._inter_ops)
+            self._inter_reqs += self._commit_ops(self._inter_ops)
             self._inter_ops = []
 
     def wait(self, force_wait_inter=False):
