@@ -1,0 +1,260 @@
+// MFMA-pipe worksheet for the round-3 asm decision.
+//
+// Question: does a 1-wave/SIMD hand-ordered stream (counted lgkmcnt,
+// reads issued AHEAD of their MFMAs) beat the production structure's
+// 2-waves/SIMD compiler schedule on the exact dependency shape of the
+// attention inner loop — chains of ds_read_b128 -> v_mfma_f32_32x32x16
+// with rotating accumulators?
+//
+// Variants (all: one 144 KB-LDS workgroup per CU so occupancy is forced;
+// 16 reads + 16 MFMAs per step, CHAINS rotating accumulators):
+//   cpp_2w : NT=512 (8 waves = 2/SIMD), compiler schedule — models the
+//            production kernels.
+//   cpp_1w : NT=256 (4 waves = 1/SIMD), compiler schedule — models the
+//            rejected BA_FWD_NT=256 structure.
+//   asm_1w : NT=256, the 16-step body in ordered inline asm with a
+//            4-deep read-ahead ring and counted lgkmcnt(3) waits.
+//
+// Build: hipcc --offload-arch=gfx950 -O3 -std=c++17 tools/mfma_pipe_probe.hip -o tools/mfma_pipe_probe
+#include <hip/hip_runtime.h>
+#include <stdio.h>
+#include <stdlib.h>
+
+typedef __attribute__((ext_vector_type(8))) _Float16 f16x8_t;
+typedef __attribute__((ext_vector_type(16))) float f32x16_t;
+typedef __attribute__((ext_vector_type(4))) unsigned int u32x4_t;
+
+#define CK(x)                                                        \
+  do {                                                               \
+    hipError_t e_ = (x);                                             \
+    if (e_ != hipSuccess) {                                          \
+      printf("HIP error %s @%d\n", hipGetErrorString(e_), __LINE__); \
+      return 1;                                                      \
+    }                                                                \
+  } while (0)
+
+constexpr int LDS_E = 72 * 1024;  // 144 KB of f16 -> one workgroup per CU
+
+template <int NT, int CHAINS>
+__global__ __launch_bounds__(NT) void pipe_cpp(float* out, int iters) {
+  __shared__ _Float16 lds[LDS_E];
+  for (int i = threadIdx.x; i < LDS_E; i += NT)
+    lds[i] = (_Float16)((i & 7) * 0.125f);
+  __syncthreads();
+  const int lane = threadIdx.x & 63;
+  f16x8_t b = *(const f16x8_t*)&lds[(lane & 31) * 8];
+  f32x16_t acc[CHAINS];
+#pragma unroll
+  for (int c = 0; c < CHAINS; ++c) acc[c] = (f32x16_t)(0.f);
+  // per-lane base, 16 B aligned, wanders per iteration (defeats hoisting)
+  int base = (threadIdx.x * 16) & (2 * LDS_E - 16);
+  for (int it = 0; it < iters; ++it) {
+#pragma unroll
+    for (int j = 0; j < 16; ++j) {
+      const int byte = (base + j * 2048) & (2 * LDS_E - 16);
+      f16x8_t a = *(const f16x8_t*)((const char*)lds + byte);
+      acc[j % CHAINS] =
+          __builtin_amdgcn_mfma_f32_32x32x16_f16(a, b, acc[j % CHAINS], 0, 0, 0);
+    }
+    base = (base + 4096) & (2 * LDS_E - 16);
+  }
+  float s = 0.f;
+#pragma unroll
+  for (int c = 0; c < CHAINS; ++c)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) s += acc[c][r];
+  if (s == 1234.5678f) out[blockIdx.x] = s;  // keep everything live
+}
+
+// softmax-dependency variant: after every 16-MFMA group, a serial VALU
+// block (max tree + 16 exp2 + packs, as the production softmax) that
+// DEPENDS on one accumulator chain and FEEDS the next group's B operand
+// — the production QK->softmax->PV structure.
+template <int NT>
+__global__ __launch_bounds__(NT) void pipe_cpp_dep(float* out, int iters) {
+  __shared__ _Float16 lds[LDS_E];
+  for (int i = threadIdx.x; i < LDS_E; i += NT)
+    lds[i] = (_Float16)((i & 7) * 0.125f);
+  __syncthreads();
+  const int lane = threadIdx.x & 63;
+  f16x8_t b = *(const f16x8_t*)&lds[(lane & 31) * 8];
+  f32x16_t acc[4];
+#pragma unroll
+  for (int c = 0; c < 4; ++c) acc[c] = (f32x16_t)(0.f);
+  int base = (threadIdx.x * 16) & (2 * LDS_E - 16);
+  for (int it = 0; it < iters; ++it) {
+#pragma unroll
+    for (int j = 0; j < 16; ++j) {
+      const int byte = (base + j * 2048) & (2 * LDS_E - 16);
+      f16x8_t a = *(const f16x8_t*)((const char*)lds + byte);
+      acc[j % 4] =
+          __builtin_amdgcn_mfma_f32_32x32x16_f16(a, b, acc[j % 4], 0, 0, 0);
+    }
+    // softmax-shaped serial VALU on chain 0, feeding b (the dependency)
+    float m = acc[0][0];
+#pragma unroll
+    for (int r = 1; r < 16; ++r) m = fmaxf(m, acc[0][r]);
+    m = fmaxf(m, __shfl_xor(m, 32));
+    float rs = 0.f;
+    _Float16 h[8];
+#pragma unroll
+    for (int r = 0; r < 8; ++r) {
+      const float e0 = __builtin_amdgcn_exp2f(acc[0][2 * r] - m);
+      const float e1 = __builtin_amdgcn_exp2f(acc[0][2 * r + 1] - m);
+      rs += e0 + e1;
+      h[r] = (_Float16)(e0 * 0.001f);
+    }
+    rs += __shfl_xor(rs, 32);
+    b = *(f16x8_t*)h;  // feeds the next group's MFMAs
+    acc[0] = (f32x16_t)(rs * 1e-30f);  // chain restarts (like st per subtile)
+    base = (base + 4096) & (2 * LDS_E - 16);
+  }
+  float s = 0.f;
+#pragma unroll
+  for (int c = 0; c < 4; ++c)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) s += acc[c][r];
+  if (s == 1234.5678f) out[blockIdx.x] = s;
+}
+
+// asm variant: 4 accumulator chains, 4-slot read ring, counted waits.
+// The whole 16-step body is ONE ordered asm block; the compiler only
+// allocates registers.
+__global__ __launch_bounds__(256) void pipe_asm(float* out, int iters) {
+  __shared__ _Float16 lds[LDS_E];
+  for (int i = threadIdx.x; i < LDS_E; i += 256)
+    lds[i] = (_Float16)((i & 7) * 0.125f);
+  __syncthreads();
+  const int lane = threadIdx.x & 63;
+  f16x8_t b = *(const f16x8_t*)&lds[(lane & 31) * 8];
+  f32x16_t a0 = (f32x16_t)(0.f), a1 = (f32x16_t)(0.f), a2 = (f32x16_t)(0.f),
+           a3 = (f32x16_t)(0.f);
+  f16x8_t r0, r1, r2, r3;
+  // 4 wandering addresses (byte offsets into the LDS object)
+  int p0 = (threadIdx.x * 16) & (2 * LDS_E - 16);
+  int p1 = (p0 + 2048) & (2 * LDS_E - 16);
+  int p2 = (p0 + 4096) & (2 * LDS_E - 16);
+  int p3 = (p0 + 6144) & (2 * LDS_E - 16);
+  // LDS base as a flat-cast pointer for ds_read addressing
+  auto lp = (__attribute__((address_space(3))) char*)lds;
+  // prologue: fill the 4-deep ring
+  asm volatile(
+      "ds_read_b128 %0, %4\n\t"
+      "ds_read_b128 %1, %5\n\t"
+      "ds_read_b128 %2, %6\n\t"
+      "ds_read_b128 %3, %7\n\t"
+      : "=v"(r0), "=v"(r1), "=v"(r2), "=v"(r3)
+      : "v"(lp + p0), "v"(lp + p1), "v"(lp + p2), "v"(lp + p3));
+  for (int it = 0; it < iters; ++it) {
+    // steady state: wait the OLDEST read only (3 younger stay in
+    // flight), MFMA it, re-issue its slot 4 ahead.  16 steps = 4 full
+    // ring turns; accumulator chain c = slot index (4 chains).
+    asm volatile(
+        "s_waitcnt lgkmcnt(3)\n\t"
+        "v_mfma_f32_32x32x16_f16 %0, %4, %8, %0\n\t"
+        "ds_read_b128 %4, %9 offset:512\n\t"
+        "s_waitcnt lgkmcnt(3)\n\t"
+        "v_mfma_f32_32x32x16_f16 %1, %5, %8, %1\n\t"
+        "ds_read_b128 %5, %10 offset:512\n\t"
+        "s_waitcnt lgkmcnt(3)\n\t"
+        "v_mfma_f32_32x32x16_f16 %2, %6, %8, %2\n\t"
+        "ds_read_b128 %6, %11 offset:512\n\t"
+        "s_waitcnt lgkmcnt(3)\n\t"
+        "v_mfma_f32_32x32x16_f16 %3, %7, %8, %3\n\t"
+        "ds_read_b128 %7, %12 offset:512\n\t"
+
+        "s_waitcnt lgkmcnt(3)\n\t"
+        "v_mfma_f32_32x32x16_f16 %0, %4, %8, %0\n\t"
+        "ds_read_b128 %4, %9 offset:1024\n\t"
+        "s_waitcnt lgkmcnt(3)\n\t"
+        "v_mfma_f32_32x32x16_f16 %1, %5, %8, %1\n\t"
+        "ds_read_b128 %5, %10 offset:1024\n\t"
+        "s_waitcnt lgkmcnt(3)\n\t"
+        "v_mfma_f32_32x32x16_f16 %2, %6, %8, %2\n\t"
+        "ds_read_b128 %6, %11 offset:1024\n\t"
+        "s_waitcnt lgkmcnt(3)\n\t"
+        "v_mfma_f32_32x32x16_f16 %3, %7, %8, %3\n\t"
+        "ds_read_b128 %7, %12 offset:1024\n\t"
+
+        "s_waitcnt lgkmcnt(3)\n\t"
+        "v_mfma_f32_32x32x16_f16 %0, %4, %8, %0\n\t"
+        "ds_read_b128 %4, %9 offset:1536\n\t"
+        "s_waitcnt lgkmcnt(3)\n\t"
+        "v_mfma_f32_32x32x16_f16 %1, %5, %8, %1\n\t"
+        "ds_read_b128 %5, %10 offset:1536\n\t"
+        "s_waitcnt lgkmcnt(3)\n\t"
+        "v_mfma_f32_32x32x16_f16 %2, %6, %8, %2\n\t"
+        "ds_read_b128 %6, %11 offset:1536\n\t"
+        "s_waitcnt lgkmcnt(3)\n\t"
+        "v_mfma_f32_32x32x16_f16 %3, %7, %8, %3\n\t"
+        "ds_read_b128 %7, %12 offset:1536\n\t"
+
+        "s_waitcnt lgkmcnt(3)\n\t"
+        "v_mfma_f32_32x32x16_f16 %0, %4, %8, %0\n\t"
+        "ds_read_b128 %4, %9\n\t"
+        "s_waitcnt lgkmcnt(3)\n\t"
+        "v_mfma_f32_32x32x16_f16 %1, %5, %8, %1\n\t"
+        "ds_read_b128 %5, %10\n\t"
+        "s_waitcnt lgkmcnt(3)\n\t"
+        "v_mfma_f32_32x32x16_f16 %2, %6, %8, %2\n\t"
+        "ds_read_b128 %6, %11\n\t"
+        "s_waitcnt lgkmcnt(3)\n\t"
+        "v_mfma_f32_32x32x16_f16 %3, %7, %8, %3\n\t"
+        "ds_read_b128 %7, %12\n\t"
+        "s_nop 7\n\t"
+        : "+v"(a0), "+v"(a1), "+v"(a2), "+v"(a3), "+v"(r0), "+v"(r1),
+          "+v"(r2), "+v"(r3)
+        : "v"(b), "v"(lp + p0), "v"(lp + p1), "v"(lp + p2), "v"(lp + p3));
+  }
+  asm volatile("s_waitcnt lgkmcnt(0)\n\ts_nop 7" ::: "memory");
+  float s = 0.f;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) s += a0[r] + a1[r] + a2[r] + a3[r];
+  // keep the ring reads live too
+  s += (float)r0[0] + (float)r1[0] + (float)r2[0] + (float)r3[0];
+  if (s == 1234.5678f) out[blockIdx.x] = s;
+}
+
+template <typename K>
+static float bench(const char* name, K kern, int nt, float* out, int iters,
+                   int waves_per_wg, int mfmas_per_iter) {
+  hipEvent_t e0, e1;
+  (void)hipEventCreate(&e0);
+  (void)hipEventCreate(&e1);
+  const int blocks = 256;  // one per CU
+  hipLaunchKernelGGL(kern, dim3(blocks), dim3(nt), 0, 0, out, iters);
+  (void)hipDeviceSynchronize();
+  hipError_t le = hipGetLastError();
+  if (le != hipSuccess) {
+    printf("%-10s LAUNCH ERROR: %s\n", name, hipGetErrorString(le));
+    return 0;
+  }
+  (void)hipEventRecord(e0);
+  hipLaunchKernelGGL(kern, dim3(blocks), dim3(nt), 0, 0, out, iters);
+  (void)hipEventRecord(e1);
+  (void)hipEventSynchronize(e1);
+  float ms = 0;
+  (void)hipEventElapsedTime(&ms, e0, e1);
+  // FLOPs: per wave-step MFMA = 2*32*32*16
+  double fl = (double)blocks * waves_per_wg * iters * mfmas_per_iter * 2.0 *
+              32 * 32 * 16;
+  double tf = fl / (ms / 1e3) / 1e12;
+  printf("%-10s %8.2f ms  %8.1f TF (%4.1f%% of 2.5PF)\n", name, ms, tf,
+         tf / 2500 * 100);
+  return tf;
+}
+
+int main(int argc, char** argv) {
+  const int iters = (argc > 1) ? atoi(argv[1]) : 200000;
+  float* out;
+  CK(hipMalloc(&out, 1024 * 4));
+  bench("cpp_2w_c1", pipe_cpp<512, 1>, 512, out, iters, 8, 16);
+  bench("cpp_2w_c2", pipe_cpp<512, 2>, 512, out, iters, 8, 16);
+  bench("cpp_2w_c4", pipe_cpp<512, 4>, 512, out, iters, 8, 16);
+  bench("cpp_1w_c4", pipe_cpp<256, 4>, 256, out, iters, 4, 16);
+  bench("cpp_1w_c8", pipe_cpp<256, 8>, 256, out, iters, 4, 16);
+  bench("cpp_2w_dep", pipe_cpp_dep<512>, 512, out, iters, 8, 16);
+  bench("cpp_1w_dep", pipe_cpp_dep<256>, 256, out, iters, 4, 16);
+  bench("asm_1w_c4", pipe_asm, 256, out, iters, 4, 16);
+  return 0;
+}
