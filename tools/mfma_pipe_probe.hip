@@ -215,6 +215,75 @@ __global__ __launch_bounds__(256) void pipe_asm(float* out, int iters) {
   if (s == 1234.5678f) out[blockIdx.x] = s;
 }
 
+// asm + gap-filler variant: same 16-MFMA/16-read ring, but each MFMA gap
+// carries 4 softmax-shaped VALU instructions (max3 / exp2 / fma / pack)
+// on a side register set — 64 VALU per group, the production softmax
+// volume.  Measures whether softmax WORK rides the 1-wave MFMA stream.
+__global__ __launch_bounds__(256) void pipe_asm_sm(float* out, int iters) {
+  __shared__ _Float16 lds[LDS_E];
+  for (int i = threadIdx.x; i < LDS_E; i += 256)
+    lds[i] = (_Float16)((i & 7) * 0.125f);
+  __syncthreads();
+  const int lane = threadIdx.x & 63;
+  f16x8_t b = *(const f16x8_t*)&lds[(lane & 31) * 8];
+  f32x16_t a0 = (f32x16_t)(0.f), a1 = (f32x16_t)(0.f), a2 = (f32x16_t)(0.f),
+           a3 = (f32x16_t)(0.f);
+  f16x8_t r0, r1, r2, r3;
+  float s0 = 0.1f * lane, s1 = 0.2f, s2 = 0.3f, s3 = 0.4f;
+  int p0 = (threadIdx.x * 16) & (2 * LDS_E - 16);
+  int p1 = (p0 + 2048) & (2 * LDS_E - 16);
+  int p2 = (p0 + 4096) & (2 * LDS_E - 16);
+  int p3 = (p0 + 6144) & (2 * LDS_E - 16);
+  auto lp = (__attribute__((address_space(3))) char*)lds;
+  asm volatile(
+      "ds_read_b128 %0, %4\n\t"
+      "ds_read_b128 %1, %5\n\t"
+      "ds_read_b128 %2, %6\n\t"
+      "ds_read_b128 %3, %7\n\t"
+      : "=v"(r0), "=v"(r1), "=v"(r2), "=v"(r3)
+      : "v"(lp + p0), "v"(lp + p1), "v"(lp + p2), "v"(lp + p3));
+  for (int it = 0; it < iters; ++it) {
+    // one macro step = wait, MFMA, read re-issue, 4 VALU fillers
+#define STEP(ACC, RD, AD, OFF)                          \
+  "s_waitcnt lgkmcnt(3)\n\t"                            \
+  "v_mfma_f32_32x32x16_f16 " ACC ", " RD ", %12, " ACC  \
+  "\n\t"                                                \
+  "ds_read_b128 " RD ", " AD " offset:" OFF "\n\t"      \
+  "v_max3_f32 %8, %8, %9, %10\n\t"                      \
+  "v_exp_f32 %9, %9\n\t"                                \
+  "v_fma_f32 %10, %10, %11, %8\n\t"                     \
+  "v_cvt_pk_f16_f32 %11, %10, %9\n\t"
+    asm volatile(
+        STEP("%0", "%4", "%13", "512")
+        STEP("%1", "%5", "%14", "512")
+        STEP("%2", "%6", "%15", "512")
+        STEP("%3", "%7", "%16", "512")
+        STEP("%0", "%4", "%13", "1024")
+        STEP("%1", "%5", "%14", "1024")
+        STEP("%2", "%6", "%15", "1024")
+        STEP("%3", "%7", "%16", "1024")
+        STEP("%0", "%4", "%13", "1536")
+        STEP("%1", "%5", "%14", "1536")
+        STEP("%2", "%6", "%15", "1536")
+        STEP("%3", "%7", "%16", "1536")
+        STEP("%0", "%4", "%13", "0")
+        STEP("%1", "%5", "%14", "0")
+        STEP("%2", "%6", "%15", "0")
+        STEP("%3", "%7", "%16", "0")
+        "s_nop 7\n\t"
+        : "+v"(a0), "+v"(a1), "+v"(a2), "+v"(a3), "+v"(r0), "+v"(r1),
+          "+v"(r2), "+v"(r3), "+v"(s0), "+v"(s1), "+v"(s2), "+v"(s3)
+        : "v"(b), "v"(lp + p0), "v"(lp + p1), "v"(lp + p2), "v"(lp + p3));
+#undef STEP
+  }
+  asm volatile("s_waitcnt lgkmcnt(0)\n\ts_nop 7" ::: "memory");
+  float s = s0 + s1 + s2 + s3;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) s += a0[r] + a1[r] + a2[r] + a3[r];
+  s += (float)r0[0] + (float)r1[0] + (float)r2[0] + (float)r3[0];
+  if (s == 1234.5678f) out[blockIdx.x] = s;
+}
+
 template <typename K>
 static float bench(const char* name, K kern, int nt, float* out, int iters,
                    int waves_per_wg, int mfmas_per_iter) {
@@ -256,5 +325,6 @@ int main(int argc, char** argv) {
   bench("cpp_2w_dep", pipe_cpp_dep<512>, 512, out, iters, 8, 16);
   bench("cpp_1w_dep", pipe_cpp_dep<256>, 256, out, iters, 4, 16);
   bench("asm_1w_c4", pipe_asm, 256, out, iters, 4, 16);
+  bench("asm_1w_sm", pipe_asm_sm, 256, out, iters, 4, 16);
   return 0;
 }
