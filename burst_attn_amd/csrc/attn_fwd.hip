@@ -44,7 +44,15 @@ namespace {
 // rescale, exp2, pack) and PV of subtile j-1 are issued while subtile
 // j's QK MFMAs fill — an explicit two-stage software pipeline carried
 // ACROSS tile boundaries (requires NBUF=4 so the previous tile's V
-// image survives one extra subtile, and a barrier every tile)
+// image survives one extra subtile, and a barrier every tile);
+// 3 = THREE-stage pipeline at ONE WAVE PER SIMD (worksheet-driven,
+// profiles/r02/mfma_pipe_worksheet.txt): phase k issues QK(k) MFMAs,
+// the softmax of subtile k-1 (producing the P fragments), and PV(k-2)
+// — three independent chains per wave, K/V fragments prefetched a full
+// phase ahead into registers.  Requires NT=256 and NBUF=4 (128 KB LDS
+// forces one 4-wave workgroup per CU = 1 wave/SIMD, where the ~500
+// register budget holds the pipeline state).  The ot rescale of
+// softmax(k-1) is ordered AFTER PV(k-2) lands (scale consistency).
 // NT: threads per workgroup (512 = 8 waves x 1 block/CU;
 //     256 = 4 waves x 2 blocks/CU — decoupled barrier groups)
 // NBUF: LDS tile buffers. 2 = stage one tile ahead, barrier every tile.
@@ -202,6 +210,105 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
     }
   };
 
+  // ---- SUBT=3 additional state: stage 2 (P fragments of subtile k-2
+  // awaiting PV) and the register-prefetched K/V operands
+  static_assert(SUBT != 3 || (NBUF == 4 && VPATH == 0 && NT == 256),
+                "SUBT=3: NT=256 (1 wave/SIMD), NBUF=4, V^T image");
+  frag pvf[2];                     // P fragments awaiting PV
+  frag kfP[SUBT == 3 ? D / 16 : 1];  // prefetched K frags (QK of phase k)
+  frag vvP[SUBT == 3 ? D / 16 : 1];  // prefetched V rowslices (PV operands)
+  int v_kv0 = -1;                  // -1 = stage 2 empty
+
+  // one SUBT=3 pipeline phase for subtile (t, kvs) using buffer `cur3`:
+  // QK(k) on prefetched kfP -> kf prefetch for k+1 -> softmax part 1 of
+  // stP (mask+max) -> PV(k-2) on prefetched vvP -> softmax part 2
+  // (rescale ordered AFTER the PV, exp2, rowsum, pack -> pvf) -> vv
+  // prefetch for the new pf -> rotate stage 1.
+  auto phase3 = [&](int t, int kvs, int cur3, bool tile_full) {
+    // --- QK(k) on the prefetched K fragments, then re-fill them for
+    // the next subtile (WAR on kfP; a full phase of flight).
+    // NOTE (measured): pinning this cadence with per-op
+    // volatile asm does NOT help — the accumulators cross the C/asm
+    // boundary through 100-470 register copies per phase and the C
+    // softmax is not gap-packed.  The plain form below is the honest
+    // compiler-scheduled 3-stage pipeline; the hand-placed version is
+    // the round-3 full-asm body.
+    f32x16_t stQ = (f32x16_t)(0.f);
+#pragma unroll
+    for (int s = 0; s < D / 16; ++s) stQ = MT::mma(kfP[s], qf[s], stQ);
+    {
+      const int nt_t = (kvs == 0) ? t : t + 1;
+      const int nkvs = kvs ^ 1;
+      if (nt_t < nt) {
+#pragma unroll
+        for (int s = 0; s < D / 16; ++s)
+          kfP[s] = ba_ld_rowslice<T, D, SWZ_K>(ldsK(nt_t % NBUF),
+                                               nkvs * 32 + l31,
+                                               16 * s + 8 * hi);
+      }
+    }
+    // --- softmax part 1 on stP: mask + row max (no ot/m2 writes)
+    float tm = BA_NEG_BIG;
+    if (p_kv0 >= 0) {
+      if (!p_full) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int kv_g = p_kv0 + ba_crow(r, 0) + 4 * hi;
+          if (!(kv_g < Sk && (!causal || kv_g <= q_row))) stP[r] = BA_NEG_BIG;
+        }
+      }
+      tm = ba_max16(stP);
+      tm = fmaxf(tm, __shfl_xor(tm, 32));
+      tm *= c2;
+    }
+    // --- PV(k-2) on the prefetched operands
+    if (v_kv0 >= 0) {
+#pragma unroll
+      for (int dt = 0; dt < D / 32; ++dt) {
+#pragma unroll
+        for (int u = 0; u < 2; ++u)
+          ot[dt] = MT::mma(vvP[dt * 2 + u], pvf[u], ot[dt]);
+      }
+    }
+    // --- softmax part 2: rescale (safe now: the PV above has landed),
+    // exp2, rowsum, pack; then prefetch the new pf's V operands
+    if (p_kv0 >= 0) {
+      if (!__all(tm - m2 <= DEFER_THR2)) {
+        const float mnew = fmaxf(m2, tm);
+        const float alpha = ba_exp2(m2 - mnew);
+        m2 = mnew;
+        lsum *= alpha;
+#pragma unroll
+        for (int dt = 0; dt < D / 32; ++dt)
+#pragma unroll
+          for (int r = 0; r < 16; ++r) ot[dt][r] *= alpha;
+      }
+      float rowsum = 0.f;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        stP[r] = ba_exp2(__builtin_fmaf(stP[r], c2, -m2));
+        rowsum += stP[r];
+      }
+      rowsum += __shfl_xor(rowsum, 32);
+      lsum += rowsum;
+      ba_build_frag_pair<T>(stP, pvf);
+      v_kv0 = p_kv0;
+      const int kvs_l = (p_kv0 / 32) & 1;
+#pragma unroll
+      for (int dt = 0; dt < D / 32; ++dt) {
+#pragma unroll
+        for (int u = 0; u < 2; ++u)
+          vvP[dt * 2 + u] = ba_ld_rowslice<T, KVBLK, SWZ_V, 7>(
+              ldsVT(p_cur), dt * 32 + l31, kvs_l * 32 + 16 * u + 8 * hi);
+      }
+    }
+    // --- rotate stage 1
+    stP = stQ;
+    p_kv0 = t * KVBLK + kvs * 32;
+    p_cur = cur3;
+    p_full = tile_full;
+  };
+
   auto issue_loads = [&](int tile, u32x4_t* kreg, u32x4_t* vreg) {
     const int kv0 = tile * KVBLK;
 #pragma unroll
@@ -268,6 +375,12 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
         kfr[0][s2] =
             ba_ld_rowslice<T, D, SWZ_K>(ldsK(0), l31, 16 * s2 + 8 * hi);
     }
+    if (SUBT == 3) {  // prime the 3-stage pipeline's K operands
+#pragma unroll
+      for (int s2 = 0; s2 < D / 16; ++s2)
+        kfP[s2] =
+            ba_ld_rowslice<T, D, SWZ_K>(ldsK(0), l31, 16 * s2 + 8 * hi);
+    }
   }
   // static priority for the younger dispatch half (T5 static form):
   // wave-uniform condition via readfirstlane, one s_setprio, no flips
@@ -283,7 +396,12 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
     if (has_next) issue_loads(t + AHEAD, kreg, vreg);
 
     const bool active = !causal || (kv0 <= qb + 31);
-    if (active && SUBT == 2) {
+    if (active && SUBT == 3) {
+      const bool tile_full3 =
+          (kv0 + KVBLK <= Sk) && (!causal || (kv0 + KVBLK - 1 <= qb));
+      phase3(t, 0, cur, tile_full3);
+      phase3(t, 1, cur, tile_full3);
+    } else if (active && SUBT == 2) {
       // ---- T15 pipeline: QK(j) fills while FINISH+PV(j-1) retire
       const bool tile_full =
           (kv0 + KVBLK <= Sk) && (!causal || (kv0 + KVBLK - 1 <= qb));
@@ -476,10 +594,21 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
     if (has_next) write_lds((t + AHEAD) % NBUF, kreg, vreg);
     // SUBT=2 reads the previous tile's V one subtile late: barrier every
     // tile so the rewrite (2 buffers ahead) never crosses those reads
-    if (SUBT == 2 || KREG == 2 || NBUF == 2 || (t & 1) || t + 1 >= nt)
+    if (SUBT == 2 || SUBT == 3 || KREG == 2 || NBUF == 2 || (t & 1) ||
+        t + 1 >= nt)
       __syncthreads();
   }
   if (SUBT == 2 && p_kv0 >= 0) finish_subtile();  // drain the pipeline
+  if (SUBT == 3) {  // drain both pending stages (PV first: scale order)
+    if (v_kv0 >= 0) {
+#pragma unroll
+      for (int dt = 0; dt < D / 32; ++dt)
+#pragma unroll
+        for (int u = 0; u < 2; ++u)
+          ot[dt] = MT::mma(vvP[dt * 2 + u], pvf[u], ot[dt]);
+    }
+    if (p_kv0 >= 0) finish_subtile();
+  }
 
   // ---- epilogue
   if (q_row < Sq) {
@@ -629,6 +758,17 @@ static int launch_fwd(const void* q, const void* k, const void* v, float* o,
     BA_CHECK_LAUNCH();
     return 0;
   }
+  if (subt == 3) {
+    attn_fwd_kernel<T, D, 64, 0, 0, 3, 256, 4>
+        <<<dim3((unsigned)((Sq + 127) / 128), (unsigned)N, (unsigned)B),
+           dim3(256), 0, (hipStream_t)stream>>>(
+            (const T*)q, (const T*)k, (const T*)v, o, lse, (int)Sq, (int)Sk,
+            (int)N, qs[0], qs[1], qs[2], ks[0], ks[1], ks[2], vs[0], vs[1],
+            vs[2], scale, causal, nullptr, nullptr, nullptr, 0, 0, 0, 0, 0,
+            0);
+    BA_CHECK_LAUNCH();
+    return 0;
+  }
   if (subt == 2) {
     attn_fwd_kernel<T, D, 64, 0, 0, 2, 512, 4>
         <<<dim3((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B),
@@ -724,6 +864,17 @@ static int launch_fwd_accum(const void* q, const void* k, const void* v,
               (int)Sq, (int)Sk, (int)N, qs[0], qs[1], qs[2], ks[0], ks[1],
               ks[2], vs[0], vs[1], vs[2], scale, causal, acc, m, l, as[0],
               as[1], as[2], mls[0], mls[1], carry_in);
+    BA_CHECK_LAUNCH();
+    return 0;
+  }
+  if (subt == 3) {
+    dim3 grid3((unsigned)((Sq + 127) / 128), (unsigned)N, (unsigned)B);
+    attn_fwd_kernel<T, D, 64, 1, 0, 3, 256, 4>
+        <<<grid3, dim3(256), 0, (hipStream_t)stream>>>(
+            (const T*)q, (const T*)k, (const T*)v, nullptr, nullptr, (int)Sq,
+            (int)Sk, (int)N, qs[0], qs[1], qs[2], ks[0], ks[1], ks[2], vs[0],
+            vs[1], vs[2], scale, causal, acc, m, l, as[0], as[1], as[2],
+            mls[0], mls[1], carry_in);
     BA_CHECK_LAUNCH();
     return 0;
   }

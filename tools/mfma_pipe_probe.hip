@@ -284,6 +284,91 @@ __global__ __launch_bounds__(256) void pipe_asm_sm(float* out, int iters) {
   if (s == 1234.5678f) out[blockIdx.x] = s;
 }
 
+// attention-shaped asm variant: the REAL phase schedule of the planned
+// asm forward.  Per phase: 8 QK MFMAs on a SINGLE accumulator chain
+// alternating with 8 PV MFMAs on 4 rotating chains (so the single
+// chain is never back-to-back with itself); 3 softmax-model VALU per
+// gap ending in the P packs; K and V each stream through a 4-slot read
+// ring at a steady counted lgkmcnt(7).  Two phases unrolled so the
+// stQ/stP states ping-pong positionally.  (The P->PV data link is
+// order-guaranteed by the stream; its value flows through a fixed
+// fragment here — the timing shape is what is measured.)
+__global__ __launch_bounds__(256) void pipe_asm_att(float* out, int iters) {
+  __shared__ _Float16 lds[LDS_E];
+  for (int i = threadIdx.x; i < LDS_E; i += 256)
+    lds[i] = (_Float16)((i & 7) * 0.125f);
+  __syncthreads();
+  const int lane = threadIdx.x & 63;
+  f16x8_t qb = *(const f16x8_t*)&lds[(lane & 31) * 8];
+  f16x8_t pb = *(const f16x8_t*)&lds[(lane & 31) * 8 + 256];
+  f32x16_t o0 = (f32x16_t)(0.f), o1 = (f32x16_t)(0.f), o2 = (f32x16_t)(0.f),
+           o3 = (f32x16_t)(0.f);
+  f32x16_t sx = (f32x16_t)(0.f), sy = (f32x16_t)(0.f);
+  f16x8_t kr0, kr1, kr2, kr3, vr0, vr1, vr2, vr3;
+  float t0 = 0.1f * lane, t1 = 0.2f, t2 = 0.3f, t3 = 0.4f;
+  int pk = (threadIdx.x * 16) & (2 * LDS_E - 16);
+  int pv = (pk + 8192) & (2 * LDS_E - 16);
+  auto lp = (__attribute__((address_space(3))) char*)lds;
+  asm volatile(  // prologue: fill both rings (interleaved K,V order)
+      "ds_read_b128 %0, %8\n\t"
+      "ds_read_b128 %4, %9\n\t"
+      "ds_read_b128 %1, %8 offset:512\n\t"
+      "ds_read_b128 %5, %9 offset:512\n\t"
+      "ds_read_b128 %2, %8 offset:1024\n\t"
+      "ds_read_b128 %6, %9 offset:1024\n\t"
+      "ds_read_b128 %3, %8 offset:1536\n\t"
+      "ds_read_b128 %7, %9 offset:1536\n\t"
+      : "=v"(kr0), "=v"(kr1), "=v"(kr2), "=v"(kr3), "=v"(vr0), "=v"(vr1),
+        "=v"(vr2), "=v"(vr3)
+      : "v"(lp + pk), "v"(lp + pv));
+  for (int it = 0; it < iters; ++it) {
+// one step pair: QK MFMA on the phase chain ST + PV MFMA on rotating OT
+#define PH_STEP(ST, OT, KR, VR, KOFF, VOFF)            \
+  "s_waitcnt lgkmcnt(7)\n\t"                           \
+  "v_mfma_f32_32x32x16_f16 " ST ", " KR ", %18, " ST   \
+  "\n\t"                                               \
+  "ds_read_b128 " KR ", %20 offset:" KOFF "\n\t"       \
+  "v_max3_f32 %14, %14, %15, %16\n\t"                  \
+  "v_exp_f32 %15, %15\n\t"                             \
+  "v_fma_f32 %16, %16, %17, %14\n\t"                   \
+  "s_waitcnt lgkmcnt(7)\n\t"                           \
+  "v_mfma_f32_32x32x16_f16 " OT ", " VR ", %19, " OT   \
+  "\n\t"                                               \
+  "ds_read_b128 " VR ", %21 offset:" VOFF "\n\t"       \
+  "v_max3_f32 %17, %17, %14, %15\n\t"                  \
+  "v_exp_f32 %14, %14\n\t"                             \
+  "v_add_f32 %15, %15, %16\n\t"
+#define PHASE(ST)                                           \
+  PH_STEP(ST, "%0", "%6", "%10", "512", "2560")             \
+  PH_STEP(ST, "%1", "%7", "%11", "1024", "3072")            \
+  PH_STEP(ST, "%2", "%8", "%12", "1536", "3584")            \
+  PH_STEP(ST, "%3", "%9", "%13", "2048", "4096")            \
+  PH_STEP(ST, "%0", "%6", "%10", "0", "2048")               \
+  PH_STEP(ST, "%1", "%7", "%11", "512", "2560")             \
+  PH_STEP(ST, "%2", "%8", "%12", "1024", "3072")            \
+  PH_STEP(ST, "%3", "%9", "%13", "1536", "3584")            \
+  "v_cvt_pk_f16_f32 %14, %14, %15\n\t"                     \
+  "v_cvt_pk_f16_f32 %16, %16, %17\n\t"
+    asm volatile(
+        PHASE("%4")
+        PHASE("%5")
+        "s_nop 7\n\t"
+        : "+v"(o0), "+v"(o1), "+v"(o2), "+v"(o3), "+v"(sx), "+v"(sy),
+          "+v"(kr0), "+v"(kr1), "+v"(kr2), "+v"(kr3), "+v"(vr0), "+v"(vr1),
+          "+v"(vr2), "+v"(vr3), "+v"(t0), "+v"(t1), "+v"(t2), "+v"(t3)
+        : "v"(qb), "v"(pb), "v"(lp + pk), "v"(lp + pv));
+#undef PHASE
+#undef PH_STEP
+  }
+  asm volatile("s_waitcnt lgkmcnt(0)\n\ts_nop 7" ::: "memory");
+  float s = t0 + t1 + t2 + t3;
+#pragma unroll
+  for (int r = 0; r < 16; ++r)
+    s += o0[r] + o1[r] + o2[r] + o3[r] + sx[r] + sy[r];
+  s += (float)kr0[0] + (float)vr0[0];
+  if (s == 1234.5678f) out[blockIdx.x] = s;
+}
+
 template <typename K>
 static float bench(const char* name, K kern, int nt, float* out, int iters,
                    int waves_per_wg, int mfmas_per_iter) {
@@ -326,5 +411,6 @@ int main(int argc, char** argv) {
   bench("cpp_1w_dep", pipe_cpp_dep<256>, 256, out, iters, 4, 16);
   bench("asm_1w_c4", pipe_asm, 256, out, iters, 4, 16);
   bench("asm_1w_sm", pipe_asm_sm, 256, out, iters, 4, 16);
+  bench("asm_1w_att", pipe_asm_att, 256, out, iters, 4, 32);
   return 0;
 }
