@@ -224,7 +224,11 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
   // stP (mask+max) -> PV(k-2) on prefetched vvP -> softmax part 2
   // (rescale ordered AFTER the PV, exp2, rowsum, pack -> pvf) -> vv
   // prefetch for the new pf -> rotate stage 1.
-  auto phase3 = [&](int t, int kvs, int cur3, bool tile_full) {
+  // `steady` folds the pipeline-occupancy guards away at the two
+  // steady-state call sites (t >= 1: both stages provably occupied) —
+  // round-3 prerequisite: the steady loop must be straight-line for the
+  // hand-scheduled .s body
+  auto phase3 = [&](bool steady, int t, int kvs, int cur3, bool tile_full) {
     // --- QK(k) on the prefetched K fragments, then re-fill them for
     // the next subtile (WAR on kfP; a full phase of flight).
     // NOTE (measured): pinning this cadence with per-op
@@ -249,7 +253,7 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
     }
     // --- softmax part 1 on stP: mask + row max (no ot/m2 writes)
     float tm = BA_NEG_BIG;
-    if (p_kv0 >= 0) {
+    if (steady || p_kv0 >= 0) {
       if (!p_full) {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
@@ -262,7 +266,7 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
       tm *= c2;
     }
     // --- PV(k-2) on the prefetched operands
-    if (v_kv0 >= 0) {
+    if (steady || v_kv0 >= 0) {
 #pragma unroll
       for (int dt = 0; dt < D / 32; ++dt) {
 #pragma unroll
@@ -272,7 +276,7 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
     }
     // --- softmax part 2: rescale (safe now: the PV above has landed),
     // exp2, rowsum, pack; then prefetch the new pf's V operands
-    if (p_kv0 >= 0) {
+    if (steady || p_kv0 >= 0) {
       if (!__all(tm - m2 <= DEFER_THR2)) {
         const float mnew = fmaxf(m2, tm);
         const float alpha = ba_exp2(m2 - mnew);
@@ -399,8 +403,12 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
     if (active && SUBT == 3) {
       const bool tile_full3 =
           (kv0 + KVBLK <= Sk) && (!causal || (kv0 + KVBLK - 1 <= qb));
-      phase3(t, 0, cur, tile_full3);
-      phase3(t, 1, cur, tile_full3);
+      // NOTE: a C-level warmup peel (separate guarded/steady calls)
+      // measured -48% — four inlined phase copies bloat the loop body.
+      // The peel belongs at the .s level (round 3), where the warmup
+      // runs before the loop label.
+      phase3(false, t, 0, cur, tile_full3);
+      phase3(false, t, 1, cur, tile_full3);
     } else if (active && SUBT == 2) {
       // ---- T15 pipeline: QK(j) fills while FINISH+PV(j-1) retire
       const bool tile_full =
