@@ -81,9 +81,55 @@ def build(verbose=True, force=False):
         cmd += ["-ltorch", "-ltorch_cpu", "-ltorch_python", "-lc10",
                 "-ltorch_hip", "-lc10_hip", "-lamdhip64"]
         _run(cmd, verbose)
+    _build_asm_hsaco(verbose, force)
     if verbose:
         print(f"built {out_so}")
     return out_so
+
+
+# ---- .s -> .hsaco side-build (the round-3 asm on-ramp) ----------------
+# The forward is ALSO shipped as a hipModule-loadable code object built
+# from its own compiler-generated assembly (tools/asm_probe recipe).
+# BA_FWD_ASM=1 routes fwd_accum through it at runtime — today that
+# re-assembles the unmodified .s (a parity/latency check of the flow);
+# round 3 patches the .s steady loop with the hand schedule before the
+# assemble step.
+LLVM = os.path.join(os.path.dirname(os.path.dirname(HIPCC)), "lib", "llvm",
+                    "bin") if "rocm" in HIPCC else "/opt/rocm/lib/llvm/bin"
+
+
+def _build_asm_hsaco(verbose, force):
+    import json
+    import re
+
+    src = os.path.join(CSRC, "attn_fwd.hip")
+    s_path = os.path.join(BUILD, "attn_fwd_scaffold.s")
+    o_path = os.path.join(BUILD, "attn_fwd_scaffold.o")
+    hsaco = os.path.join(PKG_DIR, "_asm_fwd.hsaco")
+    syms = os.path.join(PKG_DIR, "_asm_fwd_syms.json")
+    if not (force or _newer(src, hsaco)):
+        return hsaco
+    _run([HIPCC, f"--offload-arch={ARCH}", "-O3", "-std=c++17", "-DNDEBUG",
+          "-S", "--cuda-device-only", src, "-o", s_path], verbose)
+    clang = os.path.join(LLVM, "clang")
+    lld = os.path.join(LLVM, "ld.lld")
+    _run([clang, "-x", "assembler", "-target", "amdgcn-amd-amdhsa",
+          f"-mcpu={ARCH}", "-c", s_path, "-o", o_path], verbose)
+    _run([lld, "-shared", o_path, "-o", hsaco], verbose)
+    # record the mangled names of the dispatchable variants (the SUBT=1
+    # NBUF=2 production instantiations, accum form, fp16 and bf16)
+    text = open(s_path).read()
+    names = {}
+    for key, pat in (
+        ("f16_accum", r"_ZN12_GLOBAL__N_115attn_fwd_kernelIDF16_Li128ELi64ELi1ELi0ELi1ELi512ELi2ELi0E\S*"),
+        ("bf16_accum", r"_ZN12_GLOBAL__N_115attn_fwd_kernelIDF16bLi128ELi64ELi1ELi0ELi1ELi512ELi2ELi0E\S*"),
+    ):
+        m = re.search(r"\.globl\t(" + pat + ")", text)
+        if m:
+            names[key] = m.group(1)
+    with open(syms, "w") as f:
+        json.dump(names, f, indent=1)
+    return hsaco
 
 
 if __name__ == "__main__":

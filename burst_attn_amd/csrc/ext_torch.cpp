@@ -234,6 +234,77 @@ std::vector<at::Tensor> attn_bwd(const at::Tensor& dout, const at::Tensor& q,
   return {dq, dk, dv};
 }
 
+// ---- hipModule side-load of the .s-built forward (round-3 on-ramp) ---
+// attn_fwd_asm_load(path, symbol): load the hsaco once per process and
+// bind the accum-form kernel; attn_fwd_accum_asm(...): same contract as
+// attn_fwd_accum, dispatched through the module (kernargs packed to the
+// kernel's exact C++ parameter layout; the HIP runtime appends the
+// hidden arguments itself — verified by tools/asm_probe).
+namespace {
+hipModule_t g_asm_mod = nullptr;
+hipFunction_t g_asm_fn[2] = {nullptr, nullptr};  // [0]=f16, [1]=bf16
+
+void attn_fwd_asm_load(const std::string& path, const std::string& sym_f16,
+                       const std::string& sym_bf16) {
+  if (g_asm_mod == nullptr)
+    TORCH_CHECK(hipModuleLoad(&g_asm_mod, path.c_str()) == hipSuccess,
+                "hipModuleLoad failed for ", path);
+  TORCH_CHECK(hipModuleGetFunction(&g_asm_fn[0], g_asm_mod,
+                                   sym_f16.c_str()) == hipSuccess,
+              "symbol not found: ", sym_f16);
+  TORCH_CHECK(hipModuleGetFunction(&g_asm_fn[1], g_asm_mod,
+                                   sym_bf16.c_str()) == hipSuccess,
+              "symbol not found: ", sym_bf16);
+}
+
+void attn_fwd_accum_asm(const at::Tensor& q, const at::Tensor& k,
+                        const at::Tensor& v, double softmax_scale,
+                        bool causal, at::Tensor& acc, at::Tensor& m,
+                        at::Tensor& l, bool carry_in) {
+  const int dt = dtype_code(q);
+  TORCH_CHECK(g_asm_fn[dt] != nullptr, "asm forward not loaded");
+  TORCH_CHECK(q.size(3) == 128, "asm forward: D=128 variants only");
+  check_qkv(q, "q");
+  check_qkv(k, "k");
+  check_qkv(v, "v");
+  check_qkv_consistent(q, k, v);
+  const auto B = q.size(0), Sq = q.size(1), N = q.size(2);
+  const auto Sk = k.size(1);
+  // exact mirror of attn_fwd_kernel's parameter list
+  struct {
+    const void *q, *k, *v;
+    float *o, *lse;
+    int Sq, Sk, N;
+    int64_t qs0, qs1, qs2, ks0, ks1, ks2, vs0, vs1, vs2;
+    float scale;
+    int causal;
+    float *st_acc, *st_m, *st_l;
+    int64_t a_sb, a_ss, a_sh, ml_sb, ml_sh;
+    int carry_in;
+  } args = {
+      q.data_ptr(), k.data_ptr(), v.data_ptr(), nullptr, nullptr,
+      (int)Sq, (int)Sk, (int)N,
+      q.stride(0), q.stride(1), q.stride(2),
+      k.stride(0), k.stride(1), k.stride(2),
+      v.stride(0), v.stride(1), v.stride(2),
+      (float)softmax_scale, causal ? 1 : 0,
+      acc.data_ptr<float>(), m.data_ptr<float>(), l.data_ptr<float>(),
+      acc.stride(0), acc.stride(1), acc.stride(2),
+      m.stride(0), m.stride(1), carry_in ? 1 : 0,
+  };
+  size_t size = sizeof(args);
+  void* cfg[] = {HIP_LAUNCH_PARAM_BUFFER_POINTER, &args,
+                 HIP_LAUNCH_PARAM_BUFFER_SIZE, &size,
+                 HIP_LAUNCH_PARAM_END};
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  TORCH_CHECK(hipModuleLaunchKernel(
+                  g_asm_fn[dt], (unsigned)((Sq + 255) / 256), (unsigned)N,
+                  (unsigned)B, 512, 1, 1, 0, stream, nullptr,
+                  cfg) == hipSuccess,
+              "asm forward launch failed");
+}
+}  // namespace
+
 at::Tensor mfma_probe(const at::Tensor& a, const at::Tensor& b) {
   TORCH_CHECK(a.is_cuda() && b.is_cuda() && a.is_contiguous() && b.is_contiguous());
   TORCH_CHECK(a.sizes() == at::IntArrayRef({32, 16}) &&
@@ -267,4 +338,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "bwd tile accumulating into fp32 dq/dk/dv views");
   m.def("mfma_probe", &mfma_probe, "32x32x16 MFMA layout probe");
   m.def("tr16_probe", &tr16_probe, "ds_read_tr16_b64 semantics probe");
+  m.def("attn_fwd_asm_load", &attn_fwd_asm_load,
+        "load the .s-built forward hsaco (round-3 on-ramp)");
+  m.def("attn_fwd_accum_asm", &attn_fwd_accum_asm,
+        "carry-in fwd via the hipModule-loaded .s kernel");
 }

@@ -55,9 +55,24 @@ class HipTileProvider:
     """gfx950 HIP kernels behind the C-ABI in include/burst_attn_hip.h."""
 
     def __init__(self):
+        import json
+        import os
+
         from . import _ext
 
         self._ext = _ext.load_extension()  # raises loudly if missing
+        # BA_FWD_ASM=1: route fwd_accum through the .s-built hsaco (the
+        # round-3 hand-scheduled-kernel on-ramp; today the unmodified
+        # re-assembly — a parity check of the module path)
+        self._asm_fwd = False
+        if os.environ.get("BA_FWD_ASM", "0") == "1":
+            here = os.path.dirname(os.path.abspath(__file__))
+            hsaco = os.path.join(here, "_asm_fwd.hsaco")
+            with open(os.path.join(here, "_asm_fwd_syms.json")) as f:
+                syms = json.load(f)
+            self._ext.attn_fwd_asm_load(hsaco, syms["f16_accum"],
+                                        syms["bf16_accum"])
+            self._asm_fwd = True
 
     def fwd(self, q, k, v, scale, causal):
         return self._ext.attn_fwd(q, k, v, float(scale), bool(causal))
@@ -85,21 +100,23 @@ class HipTileProvider:
     # running max, l fp32 [B,N,S] running sum) over the rank's FULL chunk;
     # row_offset targets the zigzag-half / striped-shift row windows.
     def fwd_accum(self, state, q, k, v, scale, causal, row_offset=0):
+        # the .s-built module carries the D=128 production variants only
+        accum = (self._ext.attn_fwd_accum_asm
+                 if self._asm_fwd and q.shape[3] == 128
+                 else self._ext.attn_fwd_accum)
         if state is None:
             assert row_offset == 0, "state is created by a full-row round"
             B, S, N, D = q.shape
             acc = torch.empty(B, S, N, D, dtype=torch.float32, device=q.device)
             m = torch.empty(B, N, S, dtype=torch.float32, device=q.device)
             l = torch.empty(B, N, S, dtype=torch.float32, device=q.device)
-            self._ext.attn_fwd_accum(q, k, v, float(scale), bool(causal),
-                                     acc, m, l, False)
+            accum(q, k, v, float(scale), bool(causal), acc, m, l, False)
             return (acc, m, l)
         acc, m, l = state
         sq = q.shape[1]
         sl = slice(row_offset, row_offset + sq)
         accv, mv, lv = acc[:, sl], m[:, :, sl], l[:, :, sl]
-        self._ext.attn_fwd_accum(q, k, v, float(scale), bool(causal),
-                                 accv, mv, lv, True)
+        accum(q, k, v, float(scale), bool(causal), accv, mv, lv, True)
         return state
 
     def fwd_finalize(self, state, out_dtype):
