@@ -107,7 +107,8 @@ def _build_asm_hsaco(verbose, force):
     o_path = os.path.join(BUILD, "attn_fwd_scaffold.o")
     hsaco = os.path.join(PKG_DIR, "_asm_fwd.hsaco")
     syms = os.path.join(PKG_DIR, "_asm_fwd_syms.json")
-    if not (force or _newer(src, hsaco)):
+    patch_mod = os.path.join(PKG_DIR, "..", "tools", "s_patch.py")
+    if not (force or _newer(src, hsaco) or _newer(patch_mod, hsaco)):
         return hsaco
     _run([HIPCC, f"--offload-arch={ARCH}", "-O3", "-std=c++17", "-DNDEBUG",
           "-S", "--cuda-device-only", src, "-o", s_path], verbose)
@@ -129,6 +130,22 @@ def _build_asm_hsaco(verbose, force):
             names[key] = m.group(1)
     with open(syms, "w") as f:
         json.dump(names, f, indent=1)
+    # hand-scheduled variant (BA_FWD_ASM=2): apply tools/s_patch.py
+    # transforms to the .s and assemble a second module
+    import importlib.util
+
+    spec = importlib.util.spec_from_file_location("s_patch", patch_mod)
+    s_patch = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(s_patch)
+    sp_path = os.path.join(BUILD, "attn_fwd_patched.s")
+    spo_path = os.path.join(BUILD, "attn_fwd_patched.o")
+    hsaco_p = os.path.join(PKG_DIR, "_asm_fwd_p.hsaco")
+    s_patch.apply("qk_split", s_path, sp_path, list(names.values()))
+    _run([clang, "-x", "assembler", "-target", "amdgcn-amd-amdhsa",
+          f"-mcpu={ARCH}", "-c", sp_path, "-o", spo_path], verbose)
+    _run([lld, "-shared", spo_path, "-o", hsaco_p], verbose)
+    if verbose:
+        print(f"built {hsaco} and {hsaco_p} (qk_split)")
     return hsaco
 
 

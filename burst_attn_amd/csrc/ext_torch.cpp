@@ -242,13 +242,17 @@ std::vector<at::Tensor> attn_bwd(const at::Tensor& dout, const at::Tensor& q,
 // hidden arguments itself — verified by tools/asm_probe).
 namespace {
 hipModule_t g_asm_mod = nullptr;
+std::string g_asm_path;
 hipFunction_t g_asm_fn[2] = {nullptr, nullptr};  // [0]=f16, [1]=bf16
 
 void attn_fwd_asm_load(const std::string& path, const std::string& sym_f16,
                        const std::string& sym_bf16) {
-  if (g_asm_mod == nullptr)
+  if (g_asm_mod == nullptr || g_asm_path != path) {
+    // earlier modules stay loaded (cheap); only the bound functions matter
     TORCH_CHECK(hipModuleLoad(&g_asm_mod, path.c_str()) == hipSuccess,
                 "hipModuleLoad failed for ", path);
+    g_asm_path = path;
+  }
   TORCH_CHECK(hipModuleGetFunction(&g_asm_fn[0], g_asm_mod,
                                    sym_f16.c_str()) == hipSuccess,
               "symbol not found: ", sym_f16);

@@ -61,16 +61,18 @@ class HipTileProvider:
         from . import _ext
 
         self._ext = _ext.load_extension()  # raises loudly if missing
-        # BA_FWD_ASM=1: route fwd_accum through the .s-built hsaco (the
-        # round-3 hand-scheduled-kernel on-ramp; today the unmodified
-        # re-assembly — a parity check of the module path)
+        # BA_FWD_ASM: route fwd_accum through a .s-built hsaco.
+        #   1 = the unmodified re-assembly (parity check of the module path)
+        #   2 = the hand-scheduled variant (tools/s_patch.py transforms)
         self._asm_fwd = False
-        if os.environ.get("BA_FWD_ASM", "0") == "1":
+        asm = os.environ.get("BA_FWD_ASM", "0")
+        if asm in ("1", "2"):
             here = os.path.dirname(os.path.abspath(__file__))
-            hsaco = os.path.join(here, "_asm_fwd.hsaco")
+            name = "_asm_fwd.hsaco" if asm == "1" else "_asm_fwd_p.hsaco"
             with open(os.path.join(here, "_asm_fwd_syms.json")) as f:
                 syms = json.load(f)
-            self._ext.attn_fwd_asm_load(hsaco, syms["f16_accum"],
+            self._ext.attn_fwd_asm_load(os.path.join(here, name),
+                                        syms["f16_accum"],
                                         syms["bf16_accum"])
             self._asm_fwd = True
 

@@ -486,3 +486,43 @@ def test_large_size_direct_parity():
     torch.testing.assert_close(dv.cpu(), dv_r, **BWD_TOL[dtype])
     torch.testing.assert_close(dk.cpu(), dk_r, **BWD_TOL[dtype])
     torch.testing.assert_close(dq.cpu(), dq_r, **BWD_TOL[dtype])
+
+
+@pytest.mark.parametrize("asm", ["1", "2"])
+@pytest.mark.parametrize("dtype", [torch.float16, torch.bfloat16])
+def test_asm_module_fwd_parity(asm, dtype, monkeypatch):
+    """The .s-assembled hipModule forward (BA_FWD_ASM) vs the in-binary
+    kernel.  asm=1 is the unmodified re-assembly and must be BITWISE
+    identical (same instructions, same launch); asm=2 is the
+    tools/s_patch.py qk_split variant, which reorders the in-D summation
+    (two partial accumulators) and gates at the tile tolerance."""
+    from burst_attn_amd.tile import HipTileProvider
+
+    base = HipTileProvider()          # plain extension dispatch
+    monkeypatch.setenv("BA_FWD_ASM", asm)
+    pa = HipTileProvider()            # module dispatch
+    assert pa._asm_fwd
+    scale = 1.0 / math.sqrt(128)
+    for b, s, n, causal, seed in [(1, 1024, 4, False, 70), (1, 744, 2, True, 71),
+                                  (2, 512, 3, False, 72)]:
+        q = _rand(b, s, n, 128, dtype, seed)
+        k = _rand(b, s, n, 128, dtype, seed + 100)
+        v = _rand(b, s, n, 128, dtype, seed + 200)
+        st_a = pa.fwd_accum(None, q, k, v, scale, causal)
+        st_b = base.fwd_accum(None, q, k, v, scale, causal)
+        if asm == "1":
+            for ta, tb in zip(st_a, st_b):
+                assert torch.equal(ta, tb), "re-assembly must be bitwise equal"
+        else:
+            o_a, lse_a = pa.fwd_finalize(st_a, dtype)
+            o_b, lse_b = base.fwd_finalize(st_b, dtype)
+            torch.testing.assert_close(o_a.float(), o_b.float(), **TOL[dtype])
+            torch.testing.assert_close(lse_a, lse_b, rtol=1e-3, atol=2e-2)
+    # D=64 falls back to the in-binary kernel rather than failing
+    q = _rand(1, 256, 2, 64, dtype, 80)
+    k = _rand(1, 256, 2, 64, dtype, 81)
+    v = _rand(1, 256, 2, 64, dtype, 82)
+    st = pa.fwd_accum(None, q, k, v, 0.125, False)
+    st_ref = base.fwd_accum(None, q, k, v, 0.125, False)
+    for ta, tb in zip(st, st_ref):
+        assert torch.equal(ta, tb)
