@@ -1,0 +1,84 @@
+"""CPU guard for the assembly patch flow (tools/s_patch.py).
+
+The round-3 on-ramp hand-edits the compiler's .s between `hipcc -S` and
+the assemble step; its anchors must fail LOUDLY if the compiler output
+drifts.  This test regenerates the .s (or uses the build tree's copy),
+applies the qk_split transform, and checks the structural invariants —
+no GPU needed (clang assembles for amdgcn without one).
+"""
+
+import json
+import os
+import re
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+BUILD = os.path.join(REPO, "burst_attn_amd", "csrc", "_build")
+S_PATH = os.path.join(BUILD, "attn_fwd_scaffold.s")
+SYMS = os.path.join(REPO, "burst_attn_amd", "_asm_fwd_syms.json")
+
+sys.path.insert(0, os.path.join(REPO, "tools"))
+import s_patch  # noqa: E402
+
+
+@pytest.fixture(scope="module")
+def scaffold():
+    if not (os.path.exists(S_PATH) and os.path.exists(SYMS)):
+        from burst_attn_amd import build_ext
+
+        build_ext._build_asm_hsaco(verbose=False, force=True)
+    return open(S_PATH).read(), json.load(open(SYMS))
+
+
+def test_symbols_extracted(scaffold):
+    _, syms = scaffold
+    assert set(syms) == {"f16_accum", "bf16_accum"}
+    for s in syms.values():
+        assert s.startswith("_ZN12_GLOBAL__N_115attn_fwd_kernel")
+
+
+def test_qk_split_structure(scaffold):
+    text, syms = scaffold
+    out = s_patch.qk_split(text, list(syms.values()))
+    for sym in syms.values():
+        body = out[out.index(sym + ":"):]
+        body = body[:body.index(".Lfunc_end")]
+        # odd chain members renamed: 8 MFMAs now target the spare range
+        spare = body.count("v_mfma_f32_32x32x16_") and len(
+            re.findall(r"v_mfma_f32_32x32x16_\w+ v\[224:239\]", body))
+        assert spare == 8, f"{sym}: expected 8 spare-accumulator MFMAs, got {spare}"
+        # merge epilogues: 16 packed adds (8 per chain)
+        adds = len(re.findall(r"v_pk_add_f32 v\[\d+:\d+\], v\[\d+:\d+\], v\[2[23]\d:2[23]\d\]", body))
+        assert adds == 16, f"{sym}: expected 16 merge adds, got {adds}"
+        # register allocation raised
+        desc = out[out.index(".amdhsa_kernel " + sym):]
+        desc = desc[:desc.index(".end_amdhsa_kernel")]
+        assert ".amdhsa_next_free_vgpr 240" in desc
+        assert ".amdhsa_accum_offset 240" in desc
+
+
+def test_qk_split_asserts_on_drift(scaffold):
+    text, syms = scaffold
+    # a scaffold without the expected chains must be rejected, not patched
+    broken = text.replace("v_mfma_f32_32x32x16_f16 v[66:81]",
+                          "v_mfma_f32_32x32x16_f16 v[40:55]")
+    with pytest.raises(AssertionError):
+        s_patch.qk_split(broken, list(syms.values()))
+
+
+def test_patched_s_assembles(scaffold, tmp_path):
+    text, syms = scaffold
+    out = s_patch.qk_split(text, list(syms.values()))
+    sp = tmp_path / "patched.s"
+    sp.write_text(out)
+    clang = "/opt/rocm/lib/llvm/bin/clang"
+    if not os.path.exists(clang):
+        pytest.skip("rocm llvm not present")
+    r = subprocess.run(
+        [clang, "-x", "assembler", "-target", "amdgcn-amd-amdhsa",
+         "-mcpu=gfx950", "-c", str(sp), "-o", str(tmp_path / "patched.o")],
+        capture_output=True, text=True)
+    assert r.returncode == 0, r.stderr[-2000:]
