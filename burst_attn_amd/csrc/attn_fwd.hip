@@ -313,8 +313,36 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
     p_full = tile_full;
   };
 
+  // staging lane offsets are loop-invariant; computing them once turns an
+  // in-range tile's addresses into (scalar tile base) + (u32 lane offset),
+  // which the compiler emits as saddr-form global_loads — the per-tile
+  // 64-bit address recompute (~32 VALU ops) leaves the steady loop, and
+  // the issue port is the oversubscribed resource there (ACTIVE+STALL
+  // ~124% of SIMD issue slots at 2 waves, profiles/r02/sq_wait_taxonomy)
+  uint32_t k_loff[PT], v_loff[PT];
+#pragma unroll
+  for (int c = 0; c < PT; ++c) {
+    const int flat = tid + c * NT;
+    const int row = flat / (D / 8);
+    const int col8 = flat % (D / 8);
+    k_loff[c] = (uint32_t)((row * k_ss + col8 * 8) * (int64_t)sizeof(T));
+    v_loff[c] = (uint32_t)((row * v_ss + col8 * 8) * (int64_t)sizeof(T));
+  }
+  const bool stage_lin =
+      (int64_t)(KVBLK - 1) * k_ss * (int64_t)sizeof(T) + 16 <= 0xffffffffLL &&
+      (int64_t)(KVBLK - 1) * v_ss * (int64_t)sizeof(T) + 16 <= 0xffffffffLL;
   auto issue_loads = [&](int tile, u32x4_t* kreg, u32x4_t* vreg) {
     const int kv0 = tile * KVBLK;
+    if (stage_lin && kv0 + KVBLK <= Sk) {  // wave-uniform: no clamp needed
+      const char* kb = (const char*)(kp + (int64_t)kv0 * k_ss);
+      const char* vb = (const char*)(vp + (int64_t)kv0 * v_ss);
+#pragma unroll
+      for (int c = 0; c < PT; ++c) {
+        if (KREG != 1) kreg[c] = *(const u32x4_t*)(kb + k_loff[c]);
+        vreg[c] = *(const u32x4_t*)(vb + v_loff[c]);
+      }
+      return;
+    }
 #pragma unroll
     for (int c = 0; c < PT; ++c) {
       const int flat = tid + c * NT;

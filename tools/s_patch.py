@@ -27,8 +27,19 @@ deterministic.
 import re
 
 ACC = (66, 81)          # compiler's QK accumulator range
-SPARE = (224, 239)      # free arch VGPRs used for the odd chain
-NEW_VGPR = 240          # next_free_vgpr / accum_offset after the rename
+
+
+def _spare_range(text, sym):
+    """First even register at/above the kernel's next_free_vgpr — the odd
+    chain's accumulator lives there so it can never collide with live
+    registers regardless of how the allocation shifts between builds."""
+    m = re.search(r"\.amdhsa_kernel\s+" + re.escape(sym) +
+                  r".*?\.amdhsa_next_free_vgpr\s+(\d+)", text, re.S)
+    assert m, f"next_free_vgpr not found for {sym}"
+    nv = int(m.group(1))
+    lo = (nv + 3) & ~3  # 4-aligned so the patched accum_offset stays legal
+    assert lo + 16 <= 256, f"{sym}: no 16-register spare below the 2-wave ceiling (nv={nv})"
+    return lo, lo + 15, lo + 16
 
 
 def _parse_mfma(line):
@@ -40,10 +51,10 @@ def _parse_mfma(line):
     return m
 
 
-def _qk_split_kernel(lines, start, end, mnem):
+def _qk_split_kernel(lines, start, end, mnem, spare_lo):
     """Apply the chain split inside one kernel's [start,end) line range."""
     acc = f"v[{ACC[0]}:{ACC[1]}]"
-    spare = f"v[{SPARE[0]}:{SPARE[1]}]"
+    spare = f"v[{spare_lo}:{spare_lo + 15}]"
     # find the accumulator-chain MFMAs, in order
     chain_idx = []
     for n in range(start, end):
@@ -71,7 +82,7 @@ def _qk_split_kernel(lines, start, end, mnem):
         for r in range(0, 16, 2):
             adds.append(
                 f"\tv_pk_add_f32 v[{ACC[0]+r}:{ACC[0]+r+1}], "
-                f"v[{ACC[0]+r}:{ACC[0]+r+1}], v[{SPARE[0]+r}:{SPARE[0]+r+1}]"
+                f"v[{ACC[0]+r}:{ACC[0]+r+1}], v[{spare_lo+r}:{spare_lo+r+1}]"
             )
         lines[ch[7]] = lines[ch[7]] + "\n" + "\n".join(adds)
     return 2
@@ -80,7 +91,7 @@ def _qk_split_kernel(lines, start, end, mnem):
 def qk_split(text, symbols):
     """Apply to every kernel named in `symbols`; returns patched text."""
     lines = text.split("\n")
-    joined_positions = {}
+    spares = {sym: _spare_range(text, sym) for sym in symbols}
     for sym in symbols:
         # kernel body range
         hdr = None
@@ -93,20 +104,20 @@ def qk_split(text, symbols):
                     if lines[n].lstrip().startswith(".Lfunc_end")
                     or lines[n].startswith(".Lfunc_end"))
         mnem = "v_mfma_f32_32x32x16_"
-        _qk_split_kernel(lines, hdr, endn, mnem)
-        joined_positions[sym] = (hdr, endn)
+        _qk_split_kernel(lines, hdr, endn, mnem, spares[sym][0])
     text = "\n".join(lines)
     # raise the register allocation for the patched kernels
     for sym in symbols:
+        new_vgpr = spares[sym][2]
         # the .amdhsa descriptor block for this kernel
         m = re.search(r"\.amdhsa_kernel\s+" + re.escape(sym), text)
         assert m, f"descriptor not found: {sym}"
         blk_end = text.index(".end_amdhsa_kernel", m.start())
         blk = text[m.start():blk_end]
-        for field, val in (("next_free_vgpr", NEW_VGPR), ("accum_offset", NEW_VGPR)):
+        for field in ("next_free_vgpr", "accum_offset"):
             blk2 = re.sub(r"(\.amdhsa_" + field + r"\s+)\d+",
-                          lambda mm: mm.group(1) + str(val), blk)
-            assert blk2 != blk or f".amdhsa_{field} {val}" in blk, \
+                          lambda mm: mm.group(1) + str(new_vgpr), blk)
+            assert blk2 != blk or f".amdhsa_{field} {new_vgpr}" in blk, \
                 f"descriptor field {field} not patched for {sym}"
             blk = blk2
         text = text[:m.start()] + blk + text[blk_end:]
@@ -114,7 +125,7 @@ def qk_split(text, symbols):
         pat = re.compile(r"(\.set\s+" + re.escape(sym) + r"\.num_vgpr,\s*)(\S+)")
         m2 = pat.search(text)
         assert m2, f".num_vgpr set-line not found: {sym}"
-        text = text[:m2.start()] + m2.group(1) + str(NEW_VGPR) + text[m2.end():]
+        text = text[:m2.start()] + m2.group(1) + str(new_vgpr) + text[m2.end():]
     return text
 
 
