@@ -73,6 +73,11 @@ namespace {
 // t+1's buffer two tiles ago and a barrier every tile orders it.  The
 // QK cluster's lgkm park (SQ_WAIT_ANY 38% in the r2 PMC taxonomy)
 // becomes overlap.  Requires SUBT=1, NBUF=4, VPATH=0.
+template <int N>
+struct ba_ic {
+  static constexpr int value = N;
+};
+
 template <typename T, int D, int KVBLK, int OUT_STATE, int VPATH, int SUBT,
           int NT = 512, int NBUF = 2, int KREG = 0>
 __global__ __launch_bounds__(NT) void attn_fwd_kernel(
@@ -420,9 +425,13 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
     __builtin_amdgcn_s_setprio(1);
 
   constexpr int AHEAD = NBUF == 4 ? 2 : 1;
-  for (int t = 0; t < nt; ++t) {
+  // the tile loop is hand-unrolled by the buffer period so `cur` (and the
+  // staging write buffer) are compile-time per copy: the LDS buffer offset
+  // folds into the ds_read/ds_write offset immediates instead of costing a
+  // v_or per access — ~40 issue slots per tile in an issue-bound loop
+  auto tile_body = [&](int t, auto curc) {
+    constexpr int cur = decltype(curc)::value;
     const int kv0 = t * KVBLK;
-    const int cur = t % NBUF;
     const bool has_next = (t + AHEAD) < nt;
     u32x4_t kreg[PT], vreg[PT];
     if (has_next) issue_loads(t + AHEAD, kreg, vreg);
@@ -633,12 +642,20 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
       }
     }
 
-    if (has_next) write_lds((t + AHEAD) % NBUF, kreg, vreg);
+    if (has_next) write_lds((cur + AHEAD) % NBUF, kreg, vreg);
     // SUBT=2 reads the previous tile's V one subtile late: barrier every
     // tile so the rewrite (2 buffers ahead) never crosses those reads
     if (SUBT == 2 || SUBT == 3 || KREG == 2 || NBUF == 2 || (t & 1) ||
         t + 1 >= nt)
       __syncthreads();
+  };
+  for (int tb = 0; tb < nt; tb += NBUF) {
+    tile_body(tb, ba_ic<0>{});
+    if (tb + 1 < nt) tile_body(tb + 1, ba_ic<1 % NBUF>{});
+    if (NBUF > 2) {
+      if (tb + 2 < nt) tile_body(tb + 2, ba_ic<2 % NBUF>{});
+      if (tb + 3 < nt) tile_body(tb + 3, ba_ic<3 % NBUF>{});
+    }
   }
   if (SUBT == 2 && p_kv0 >= 0) finish_subtile();  // drain the pipeline
   if (SUBT == 3) {  // drain both pending stages (PV first: scale order)

@@ -49,17 +49,21 @@ def test_qk_split_structure(scaffold):
         lo, hi, new_vgpr = s_patch._spare_range(text, sym)
         body = out[out.index(sym + ":"):]
         body = body[:body.index(".Lfunc_end")]
-        # odd chain members renamed: 8 MFMAs now target the spare range
+        # odd chain members renamed: 4 MFMAs per 8-deep chain target the
+        # spare range; chain count follows the hand-unrolled tile copies
         spare = len(re.findall(
             r"v_mfma_f32_32x32x16_\w+ v\[%d:%d\]" % (lo, hi), body))
-        assert spare == 8, f"{sym}: expected 8 spare-accumulator MFMAs, got {spare}"
-        # merge epilogues: 16 packed adds (8 per chain)
-        adds = len(re.findall(
-            r"v_pk_add_f32 v\[\d+:\d+\], v\[\d+:\d+\], v\[%d:\d+\]" % lo, body))
-        assert adds == 2, f"{sym}: expected 2 first-pair merge adds, got {adds}"
-        adds_all = len(re.findall(
-            r"v_pk_add_f32 v\[\d+:\d+\], v\[\d+:\d+\], v\[\d+:\d+\]", body))
-        assert adds_all == 16, f"{sym}: expected 16 merge adds, got {adds_all}"
+        assert spare >= 8 and spare % 4 == 0, \
+            f"{sym}: expected 4 spare-accumulator MFMAs per chain, got {spare}"
+        n_chains = spare // 4
+        # merge epilogues: 8 packed adds per chain, sourcing the spare
+        # range (the kernel's own rowsum also uses v_pk_add_f32 — those
+        # source softmax registers, not the spare accumulator)
+        srcs = [int(s) for s in re.findall(
+            r"v_pk_add_f32 v\[\d+:\d+\], v\[\d+:\d+\], v\[(\d+):\d+\]", body)]
+        adds_all = sum(1 for s in srcs if lo <= s <= hi)
+        assert adds_all == 8 * n_chains, \
+            f"{sym}: expected {8 * n_chains} merge adds, got {adds_all}"
         # register allocation raised past the spare range, 4-aligned
         desc = out[out.index(".amdhsa_kernel " + sym):]
         desc = desc[:desc.index(".end_amdhsa_kernel")]

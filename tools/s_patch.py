@@ -55,14 +55,17 @@ def _qk_split_kernel(lines, start, end, mnem, spare_lo):
     """Apply the chain split inside one kernel's [start,end) line range."""
     acc = f"v[{ACC[0]}:{ACC[1]}]"
     spare = f"v[{spare_lo}:{spare_lo + 15}]"
-    # find the accumulator-chain MFMAs, in order
+    # find the accumulator-chain MFMAs, in order; the steady loop holds
+    # one 8-deep chain per subtile (2 per tile copy — the tile loop is
+    # hand-unrolled by the buffer period, so 2 copies at NBUF=2)
     chain_idx = []
     for n in range(start, end):
         m = _parse_mfma(lines[n])
         if m and m.group(2).startswith(mnem) and f"v[{m.group(3)}:{m.group(4)}]" == acc:
             chain_idx.append(n)
-    assert len(chain_idx) == 16, f"expected 16 QK mfma into {acc}, got {len(chain_idx)}"
-    chains = [chain_idx[:8], chain_idx[8:]]
+    assert chain_idx and len(chain_idx) % 8 == 0, \
+        f"expected a multiple of 8 QK mfma into {acc}, got {len(chain_idx)}"
+    chains = [chain_idx[i:i + 8] for i in range(0, len(chain_idx), 8)]
     for ch in chains:
         m0 = _parse_mfma(lines[ch[0]])
         assert m0.group(7) == "0", f"chain start must accumulate from 0: {lines[ch[0]]}"
@@ -85,7 +88,7 @@ def _qk_split_kernel(lines, start, end, mnem, spare_lo):
                 f"v[{ACC[0]+r}:{ACC[0]+r+1}], v[{spare_lo+r}:{spare_lo+r+1}]"
             )
         lines[ch[7]] = lines[ch[7]] + "\n" + "\n".join(adds)
-    return 2
+    return len(chains)
 
 
 def qk_split(text, symbols):
