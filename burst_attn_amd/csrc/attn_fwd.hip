@@ -522,18 +522,12 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
 #pragma unroll
             for (int r = 0; r < 16; ++r) ot[dt][r] *= alpha;
         }
-        // rowsum as two packed lanes: the SLP vectorizer lowers the
-        // float2 accumulate to v_pk_add_f32, halving the add slots the
-        // serial form spent at the oversubscribed issue port
-        float2 rs2 = {0.f, 0.f};
+        float rowsum = 0.f;
 #pragma unroll
-        for (int r = 0; r < 16; r += 2) {
+        for (int r = 0; r < 16; ++r) {
           st[r] = ba_exp2(__builtin_fmaf(st[r], c2, -m2));
-          st[r + 1] = ba_exp2(__builtin_fmaf(st[r + 1], c2, -m2));
-          rs2.x += st[r];
-          rs2.y += st[r + 1];
+          rowsum += st[r];
         }
-        float rowsum = rs2.x + rs2.y;
         rowsum += __shfl_xor(rowsum, 32);
         lsum += rowsum;
         frag pf[2];
