@@ -77,6 +77,12 @@ template <int N>
 struct ba_ic {
   static constexpr int value = N;
 };
+struct ba_dyn {
+  int value;
+};
+template <int N>
+__device__ __forceinline__ int ba_buf_of(ba_ic<N>) { return N; }
+__device__ __forceinline__ int ba_buf_of(ba_dyn d) { return d.value; }
 
 template <typename T, int D, int KVBLK, int OUT_STATE, int VPATH, int SUBT,
           int NT = 512, int NBUF = 2, int KREG = 0>
@@ -338,7 +344,9 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
       (int64_t)(KVBLK - 1) * v_ss * (int64_t)sizeof(T) + 16 <= 0xffffffffLL;
   auto issue_loads = [&](int tile, u32x4_t* kreg, u32x4_t* vreg) {
     const int kv0 = tile * KVBLK;
-    if (stage_lin && kv0 + KVBLK <= Sk) {  // wave-uniform: no clamp needed
+    // the fast path's invariant offsets cost live registers the SUBT=2/3
+    // pipelined variants cannot spare (they spill); production SUBT=1 only
+    if (SUBT == 1 && stage_lin && kv0 + KVBLK <= Sk) {
       const char* kb = (const char*)(kp + (int64_t)kv0 * k_ss);
       const char* vb = (const char*)(vp + (int64_t)kv0 * v_ss);
 #pragma unroll
@@ -430,7 +438,7 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
   // folds into the ds_read/ds_write offset immediates instead of costing a
   // v_or per access — ~40 issue slots per tile in an issue-bound loop
   auto tile_body = [&](int t, auto curc) {
-    constexpr int cur = decltype(curc)::value;
+    const int cur = ba_buf_of(curc);  // compile-time literal for ba_ic
     const int kv0 = t * KVBLK;
     const bool has_next = (t + AHEAD) < nt;
     u32x4_t kreg[PT], vreg[PT];
@@ -444,8 +452,8 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
       // measured -48% — four inlined phase copies bloat the loop body.
       // The peel belongs at the .s level (round 3), where the warmup
       // runs before the loop label.
-      phase3(false, t, 0, cur, tile_full3);
-      phase3(false, t, 1, cur, tile_full3);
+      [[clang::always_inline]] phase3(false, t, 0, cur, tile_full3);
+      [[clang::always_inline]] phase3(false, t, 1, cur, tile_full3);
     } else if (active && SUBT == 2) {
       // ---- T15 pipeline: QK(j) fills while FINISH+PV(j-1) retire
       const bool tile_full =
@@ -459,7 +467,7 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
                                                 16 * s2 + 8 * hi);
           stQ = MT::mma(kf, qf[s2], stQ);
         }
-        if (p_kv0 >= 0) finish_subtile();
+        if (p_kv0 >= 0) [[clang::always_inline]] finish_subtile();
         stP = stQ;
         p_kv0 = kv0 + kvs * 32;
         p_cur = cur;
@@ -643,13 +651,21 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
         t + 1 >= nt)
       __syncthreads();
   };
-  for (int tb = 0; tb < nt; tb += NBUF) {
-    tile_body(tb, ba_ic<0>{});
-    if (tb + 1 < nt) tile_body(tb + 1, ba_ic<1 % NBUF>{});
-    if (NBUF > 2) {
-      if (tb + 2 < nt) tile_body(tb + 2, ba_ic<2 % NBUF>{});
-      if (tb + 3 < nt) tile_body(tb + 3, ba_ic<3 % NBUF>{});
+  if constexpr (SUBT == 1) {
+    for (int tb = 0; tb < nt; tb += NBUF) {
+      [[clang::always_inline]] tile_body(tb, ba_ic<0>{});
+      if (tb + 1 < nt) [[clang::always_inline]] tile_body(tb + 1, ba_ic<1 % NBUF>{});
+      if (NBUF > 2) {
+        if (tb + 2 < nt) [[clang::always_inline]] tile_body(tb + 2, ba_ic<2 % NBUF>{});
+        if (tb + 3 < nt) [[clang::always_inline]] tile_body(tb + 3, ba_ic<3 % NBUF>{});
+      }
     }
+  } else {
+    // the pipelined variants (SUBT=2/3) carry register state across tile
+    // boundaries; unrolled copies quadruple it into scratch spills, so
+    // they keep the rolled loop with a runtime buffer index
+    for (int t = 0; t < nt; ++t)
+      [[clang::always_inline]] tile_body(t, ba_dyn{t % NBUF});
   }
   if (SUBT == 2 && p_kv0 >= 0) finish_subtile();  // drain the pipeline
   if (SUBT == 3) {  // drain both pending stages (PV first: scale order)
