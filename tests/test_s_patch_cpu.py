@@ -94,3 +94,32 @@ def test_patched_s_assembles(scaffold, tmp_path):
          "-mcpu=gfx950", "-c", str(sp), "-o", str(tmp_path / "patched.o")],
         capture_output=True, text=True)
     assert r.returncode == 0, r.stderr[-2000:]
+
+
+def test_fwd_kernels_no_outlining_no_spills(scaffold):
+    """Regression net for the inliner: every fwd kernel instantiation must
+    be call-free (an outlined helper lambda turns into an s_swappc whose
+    ABI spills hundreds of scratch bytes — this crushed the SUBT=2/3
+    variants once) and spill-free (the production and scaffold configs
+    are tuned to their register budgets)."""
+    text, _ = scaffold
+    checked = 0
+    for m in re.finditer(r"\.amdhsa_kernel\s+(\S+)", text):
+        sym = m.group(1)
+        if "attn_fwd_kernel" not in sym:
+            continue
+        desc = text[m.start():text.index(".end_amdhsa_kernel", m.start())]
+        spill = int(re.search(
+            r"\.amdhsa_private_segment_fixed_size\s+(\d+)", desc).group(1))
+        body_start = re.search("^" + re.escape(sym) + r":.*$", text, re.M)
+        body = text[body_start.start():text.index(".Lfunc_end",
+                                                  body_start.start())]
+        assert "s_swappc" not in body, f"{sym}: outlined call in kernel body"
+        # spill-free is required for the maintained configs (SUBT=1
+        # production, SUBT=2/3 pipelined scaffolds); the legacy SUBT=0
+        # joint-softmax variant carries a pre-existing few-byte spill
+        subt = re.search(r"Li128ELi64ELi\dELi\dELi(\d)E", sym)
+        if subt and subt.group(1) != "0":
+            assert spill == 0, f"{sym}: {spill} bytes of scratch spill"
+        checked += 1
+    assert checked >= 20, f"only {checked} fwd kernels found in the scaffold"
