@@ -788,26 +788,15 @@ static int launch_fwd(const void* q, const void* k, const void* v, float* o,
                       float* lse, int64_t B, int64_t Sq, int64_t Sk, int64_t N,
                       const int64_t* qs, const int64_t* ks, const int64_t* vs,
                       float scale, int causal, void* stream) {
-  static const int vpath = [] {
-    const char* e = getenv("BA_FWD_VPATH");
-    return e ? atoi(e) : 0;  // 0 = V^T image (tr16 measured -3%; see profiles/r01)
-  }();
-  static const int subt = [] {
-    const char* e = getenv("BA_FWD_SUBT");
-    return e ? atoi(e) : 1;  // per-subtile softmax pipeline (+3% measured)
-  }();
-  static const int ntw = [] {
-    const char* e = getenv("BA_FWD_NT");
-    return e ? atoi(e) : 512;
-  }();
-  static const int nbuf = [] {
-    const char* e = getenv("BA_FWD_NBUF");
-    return e ? atoi(e) : 2;
-  }();
-  static const int kreg = [] {
-    const char* e = getenv("BA_FWD_KREG");
-    return e ? atoi(e) : 0;
-  }();
+  // read per call (getenv is ~ns against ms-scale launches) so tests can
+  // flip the variant paths in-process, like the backward plans do.
+  // defaults: V^T image (tr16 measured -3%), per-subtile softmax (+3%)
+  const char* e;
+  const int vpath = (e = getenv("BA_FWD_VPATH")) ? atoi(e) : 0;
+  const int subt = (e = getenv("BA_FWD_SUBT")) ? atoi(e) : 1;
+  const int ntw = (e = getenv("BA_FWD_NT")) ? atoi(e) : 512;
+  const int nbuf = (e = getenv("BA_FWD_NBUF")) ? atoi(e) : 2;
+  const int kreg = (e = getenv("BA_FWD_KREG")) ? atoi(e) : 0;
   if (kreg == 1 || kreg == 2) {
     dim3 gk((unsigned)((Sq + 255) / 256), (unsigned)N, (unsigned)B);
     if (kreg == 1)

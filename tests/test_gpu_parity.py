@@ -526,3 +526,34 @@ def test_asm_module_fwd_parity(asm, dtype, monkeypatch):
     st_ref = base.fwd_accum(None, q, k, v, 0.125, False)
     for ta, tb in zip(st, st_ref):
         assert torch.equal(ta, tb)
+
+
+@pytest.mark.parametrize("knobs", [
+    {"BA_FWD_SUBT": "2", "BA_FWD_NBUF": "4"},   # T15 double-pipeline
+    {"BA_FWD_SUBT": "3", "BA_FWD_NBUF": "4"},   # 3-stage 1-wave scaffold
+    {"BA_FWD_NBUF": "4"},                        # 4-buffer staging
+    {"BA_FWD_VPATH": "1"},                       # tr16 V path
+    {"BA_FWD_SUBT": "0"},                        # joint softmax
+], ids=["subt2", "subt3", "nbuf4", "vpath1", "subt0"])
+@pytest.mark.parametrize("causal", [False, True])
+@pytest.mark.parametrize("dtype", [torch.float16, torch.bfloat16])
+def test_fwd_alt_paths_vs_default(knobs, causal, dtype, monkeypatch):
+    """Every env-selectable forward variant must agree with the default
+    path (the launchers read the knobs per call, so this runs in-process
+    like the backward-plan test).  Ragged sizes included: the variants
+    share masking/boundary code but differ in staging and softmax
+    structure, where boundary bugs would hide."""
+    ext = _ext()
+    scale = 1.0 / math.sqrt(128)
+    for s, seed in ((768, 130), (744, 131)):
+        q = _rand(1, s, 2, 128, dtype, seed)
+        k = _rand(1, s, 2, 128, dtype, seed + 10)
+        v = _rand(1, s, 2, 128, dtype, seed + 20)
+        o_ref, lse_ref = ext.attn_fwd(q, k, v, scale, causal)
+        for key, val in knobs.items():
+            monkeypatch.setenv(key, val)
+        o_alt, lse_alt = ext.attn_fwd(q, k, v, scale, causal)
+        for key in knobs:
+            monkeypatch.delenv(key)
+        torch.testing.assert_close(o_alt, o_ref, **TOL[dtype])
+        torch.testing.assert_close(lse_alt, lse_ref, rtol=1e-3, atol=2e-2)
